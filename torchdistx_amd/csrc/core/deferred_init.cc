@@ -1,0 +1,687 @@
+// See deferred_init.h. Reference behavior map (file:line into
+// /root/reference/src/cc/torchdistx/deferred_init.cc):
+//   recording fallback + stack copy      — :732-862, :88-100
+//   Op / OpNode / TensorRecord tape      — :157-462
+//   view keep-alives                     — :126-154, :431-462
+//   materialization (call-stack build)   — :506-667
+//   VariableHooks proxy                  — :889-948, :1050-1128
+//
+// The tape is re-designed rather than translated: one mutable TensorRecord
+// per fake impl (restamped in place so view keep-alives follow in-place
+// rewrites), visit-order input slots instead of nulled stack entries, and a
+// single fixpoint set-collection covering the reference's three collection
+// rules (dependency closure, same-storage writers up to the last aliasing
+// writer, and readers whose inputs a later replayed writer would clobber).
+
+#include "deferred_init.h"
+
+#include <algorithm>
+#include <atomic>
+#include <functional>
+#include <memory>
+#include <mutex>
+#include <optional>
+#include <unordered_set>
+#include <vector>
+
+#include <ATen/ThreadLocalState.h>
+#include <ATen/core/dispatch/Dispatcher.h>
+#include <ATen/core/VariableHooksInterface.h>
+#include <c10/core/impl/LocalDispatchKeySet.h>
+#include <torch/library.h>
+
+#include "fake.h"
+#include "stack_utils.h"
+
+namespace tdx {
+namespace {
+
+constexpr c10::DispatchKey kDeferredKey = c10::DispatchKey::DeferredInit;
+constexpr c10::DispatchKey kFakeKey = c10::DispatchKey::Fake;
+
+struct OpNode;
+
+// Producer edge: which output of which node materializes a given fake.
+struct OpOutputDescriptor {
+  std::shared_ptr<OpNode> node;
+  size_t index = 0;  // tensor-visit-order index into the node's outputs
+};
+
+// Per-fake-tensor record, stored in the impl's dispatch-key side table.
+// Mutable: an in-place op restamps `desc` in place, so records held alive
+// through `keep_alive` (view relationships) keep following the tape.
+struct TensorRecord {
+  OpOutputDescriptor desc;
+  std::vector<std::shared_ptr<TensorRecord>> keep_alive;
+};
+
+// One recorded call frame.
+struct RecordedOp {
+  std::string name;
+  // Contract: consumes the arguments from the stack and leaves exactly the
+  // op's return values on it (OperatorHandle::callBoxed semantics).
+  std::function<void(torch::jit::Stack&)> run;
+  std::vector<c10::IValue> args;  // deep-copied; fake slots -> undefined
+  at::ThreadLocalState tls;       // captured with DeferredInit excluded
+};
+
+struct InputSlot {
+  std::optional<OpOutputDescriptor> desc;  // set when the arg was fake
+  std::optional<uint32_t> external_version;  // set for real tensor args
+};
+
+struct OpNode {
+  uint64_t op_nr = 0;
+  std::optional<RecordedOp> op;          // dropped after replay
+  std::vector<InputSlot> input_slots;    // tensor visit order over args
+  std::vector<c10::Storage> output_storages;  // meta storages, visit order
+  std::vector<std::weak_ptr<OpNode>> dependents;
+  std::vector<at::Tensor> outputs;  // real tensors after replay
+  bool materialized = false;
+};
+
+std::atomic<uint64_t> next_op_nr{1};
+
+std::shared_ptr<TensorRecord> getRecord(FakeTensorImpl* fake) {
+  return std::static_pointer_cast<TensorRecord>(fake->getData(kDeferredKey));
+}
+
+void setRecord(FakeTensorImpl* fake, std::shared_ptr<TensorRecord> rec) {
+  fake->setData(kDeferredKey, std::move(rec));
+}
+
+uint32_t tensorVersion(const at::Tensor& t) {
+  return t.unsafeGetTensorImpl()->version_counter().current_version();
+}
+
+bool storagesAlias(const c10::Storage& a, const c10::Storage& b) {
+  return a && b && a.is_alias_of(b);
+}
+
+// ---------------------------------------------------------------------------
+// Recording.
+// ---------------------------------------------------------------------------
+
+thread_local size_t deferred_level = 0;
+
+bool shouldRecordHere() {
+  return deferred_level > 0 &&
+         !c10::impl::tls_is_dispatch_key_excluded(kDeferredKey);
+}
+
+// Creates an OpNode out of a recorded call frame. `args` is the deep-copied
+// pre-execution argument region; `stack`/`rets_begin` point at the
+// post-execution return values, whose fake tensors get stamped with this
+// node's records.
+void recordOp(std::string name,
+              std::function<void(torch::jit::Stack&)> run,
+              std::vector<c10::IValue> args,
+              at::ThreadLocalState tls,
+              torch::jit::Stack& stack,
+              size_t rets_begin) {
+  auto node = std::make_shared<OpNode>();
+  node->op_nr = next_op_nr.fetch_add(1, std::memory_order_relaxed);
+
+  // Input pass: capture producer descriptors for fake args (replacing them
+  // with undefined placeholders in the saved frame) and version counters
+  // for external (real) tensor args.
+  std::vector<std::shared_ptr<TensorRecord>> input_records;
+  std::vector<c10::Storage> input_storages;
+  {
+    torch::jit::Stack args_stack{args.begin(), args.end()};
+    mapTensors(args_stack, 0, args_stack.size(),
+               [&](const at::Tensor& t) -> at::Tensor {
+                 if (auto* fake = asFake(t)) {
+                   auto rec = getRecord(fake);
+                   TORCH_CHECK(
+                       rec != nullptr,
+                       "A fake tensor constructed outside of a deferred-init "
+                       "context (or stripped of its record) was used in `",
+                       name, "` and cannot be recorded.");
+                   node->input_slots.push_back(InputSlot{rec->desc, {}});
+                   rec->desc.node->dependents.emplace_back(node);
+                   input_records.push_back(std::move(rec));
+                   input_storages.push_back(fake->meta_tensor().storage());
+                   return at::Tensor{};
+                 }
+                 if (t.defined()) {
+                   node->input_slots.push_back(
+                       InputSlot{{}, tensorVersion(t)});
+                 } else {
+                   node->input_slots.push_back(InputSlot{});
+                 }
+                 return t;
+               });
+    args.assign(args_stack.begin(), args_stack.end());
+  }
+
+  node->op = RecordedOp{std::move(name), std::move(run), std::move(args),
+                        std::move(tls)};
+
+  // Output pass: stamp (or restamp) each fake output's record.
+  visitTensors(stack, rets_begin, stack.size(), [&](const at::Tensor& t) {
+    size_t out_idx = node->output_storages.size();
+    auto* fake = asFake(t);
+    if (fake == nullptr) {
+      node->output_storages.emplace_back();
+      return;
+    }
+    const c10::Storage& out_storage = fake->meta_tensor().storage();
+    node->output_storages.push_back(out_storage);
+
+    auto rec = getRecord(fake);
+    if (rec == nullptr) {
+      rec = std::make_shared<TensorRecord>();
+      setRecord(fake, rec);
+    }
+    // Restamp IN PLACE: records reached through view keep-alives observe
+    // the rewrite (reference's view_records_, deferred_init.cc:126-154).
+    rec->desc = OpOutputDescriptor{node, out_idx};
+
+    // View keep-alive wiring: when an output aliases a *different* fake
+    // input, tie the two records' lifetimes together so whichever side
+    // outlives the other still holds the tape segments touching the shared
+    // storage.
+    for (size_t i = 0; i < input_records.size(); ++i) {
+      if (input_records[i].get() != rec.get() &&
+          storagesAlias(out_storage, input_storages[i])) {
+        rec->keep_alive.push_back(input_records[i]);
+        input_records[i]->keep_alive.push_back(rec);
+      }
+    }
+  });
+}
+
+bool isTerminalOp(const c10::OperatorHandle& op) {
+  const auto& name = op.schema().operator_name().name;
+  return name == "aten::item" || name == "aten::_local_scalar_dense";
+}
+
+void deferredInitHandler(const c10::OperatorHandle& op,
+                         c10::DispatchKeySet ks,
+                         torch::jit::Stack* stack) {
+  c10::impl::ExcludeDispatchKeyGuard no_reentry{kDeferredKey};
+
+  const auto& schema = op.schema();
+  size_t num_args = schema.arguments().size();
+  size_t args_begin = stack->size() - num_args;
+
+  bool has_fake_arg = false;
+  visitTensors(*stack, args_begin, args_begin + num_args,
+               [&](const at::Tensor& t) {
+                 if (auto* fake = asFake(t)) {
+                   has_fake_arg = true;
+                   TORCH_CHECK(
+                       getRecord(fake) != nullptr,
+                       "A fake tensor without a deferred-init record was "
+                       "passed to `", schema.operator_name(),
+                       "`. Only tensors constructed inside `deferred_init()` "
+                       "can be used here.");
+                 }
+               });
+
+  const auto below =
+      ks & c10::DispatchKeySet{c10::DispatchKeySet::FULL_AFTER, kDeferredKey};
+
+  // Terminal ops need real data (e.g. aten::item): materialize the fake
+  // arguments eagerly and run for real (reference deferred_init.cc:813-815).
+  if (has_fake_arg && isTerminalOp(op)) {
+    mapTensors(*stack, args_begin, args_begin + num_args,
+               [&](const at::Tensor& t) -> at::Tensor {
+                 return isFake(t) ? materializeTensor(t) : t;
+               });
+    op.redispatchBoxed(below, stack);
+    return;
+  }
+
+  auto saved = copyStackRegion(*stack, args_begin, args_begin + num_args);
+  at::ThreadLocalState tls;  // DeferredInit is excluded in this snapshot
+
+  // Execute through the fake layer: shape/dtype work happens on the meta
+  // backend and fake tensors come back.
+  op.redispatchBoxed(below.add(kFakeKey), stack);
+
+  size_t rets_begin = stack->size() - schema.returns().size();
+  bool has_fake_ret = false;
+  visitTensors(*stack, rets_begin, stack->size(), [&](const at::Tensor& t) {
+    has_fake_ret = has_fake_ret || isFake(t);
+  });
+
+  if (has_fake_arg || has_fake_ret) {
+    recordOp(
+        schema.operator_name().name,
+        [handle = op](torch::jit::Stack& s) { handle.callBoxed(s); },
+        std::move(saved), std::move(tls), *stack, rets_begin);
+  }
+}
+
+TORCH_LIBRARY_IMPL(_, DeferredInit, m) {
+  m.fallback(torch::CppFunction::makeFromBoxedFunction<&deferredInitHandler>());
+}
+
+// ---------------------------------------------------------------------------
+// VariableHooks proxy: `.data` reads and writes bypass the dispatcher, so a
+// recording proxy is swapped in for the duration of deferred-init
+// (reference deferred_init.cc:889-948, 1050-1128; design note in the
+// reference docs fake_tensor_and_deferred_init.rst:153-188).
+// ---------------------------------------------------------------------------
+
+class ProxyVariableHooks final : public at::impl::VariableHooksInterface {
+ public:
+  explicit ProxyVariableHooks(at::impl::VariableHooksInterface* inner)
+      : inner_{inner} {}
+
+  at::TensorBase variable_data(const at::TensorBase& self) const override;
+  void set_data(const at::TensorBase& self,
+                const at::TensorBase& new_data) const override;
+  at::TensorBase data(const at::TensorBase& self) const override {
+    // Route through variable_data so `.data` reads are recorded too.
+    return variable_data(self);
+  }
+
+  at::TensorBase tensor_data(const at::TensorBase& t) const override {
+    return inner_->tensor_data(t);
+  }
+  const std::shared_ptr<torch::autograd::Node>& grad_fn(
+      const at::TensorBase& t) const override {
+    return inner_->grad_fn(t);
+  }
+  unsigned _register_hook(
+      const at::TensorBase& t,
+      std::function<at::TensorBase(const at::TensorBase&)> hook)
+      const override {
+    return inner_->_register_hook(t, std::move(hook));
+  }
+  void remove_hook(const at::TensorBase& t, unsigned pos) const override {
+    inner_->remove_hook(t, pos);
+  }
+  bool is_view(const at::TensorBase& t) const override {
+    return inner_->is_view(t);
+  }
+  const at::TensorBase& base(const at::TensorBase& t) const override {
+    return inner_->base(t);
+  }
+  const std::string& name(const at::TensorBase& t) const override {
+    return inner_->name(t);
+  }
+  bool is_leaf(const at::TensorBase& t) const override {
+    return inner_->is_leaf(t);
+  }
+  int64_t output_nr(const at::TensorBase& t) const override {
+    return inner_->output_nr(t);
+  }
+  int64_t _version(const at::TensorBase& t) const override {
+    return inner_->_version(t);
+  }
+  void retain_grad(const at::TensorBase& t) const override {
+    inner_->retain_grad(t);
+  }
+  bool retains_grad(const at::TensorBase& t) const override {
+    return inner_->retains_grad(t);
+  }
+  void _backward(const at::Tensor& t,
+                 at::TensorList inputs,
+                 const std::optional<at::Tensor>& gradient,
+                 std::optional<bool> keep_graph,
+                 bool create_graph) const override {
+    inner_->_backward(t, inputs, gradient, keep_graph, create_graph);
+  }
+  void requires_grad_(const at::TensorBase& t, bool value) const override {
+    inner_->requires_grad_(t, value);
+  }
+  void basic_autograd_not_implemented_fallback(
+      const c10::OperatorHandle& op,
+      c10::DispatchKeySet dispatch_keys,
+      torch::jit::Stack* stack) const override {
+    inner_->basic_autograd_not_implemented_fallback(op, dispatch_keys, stack);
+  }
+  std::optional<c10::ScalarType> grad_dtype(
+      const at::TensorBase& t) const override {
+    return inner_->grad_dtype(t);
+  }
+  void set_grad_dtype(
+      const at::TensorBase& t,
+      const std::optional<c10::ScalarType>& dtype) const override {
+    inner_->set_grad_dtype(t, dtype);
+  }
+
+  at::impl::VariableHooksInterface* inner() const noexcept {
+    return inner_;
+  }
+
+ private:
+  at::impl::VariableHooksInterface* inner_;
+};
+
+std::mutex hooks_mutex;
+size_t hooks_refcount = 0;
+at::impl::VariableHooksInterface* saved_hooks = nullptr;
+ProxyVariableHooks* proxy_hooks = nullptr;
+
+void installProxyHooks() {
+  std::lock_guard<std::mutex> lock{hooks_mutex};
+  if (hooks_refcount++ == 0) {
+    saved_hooks = at::impl::GetVariableHooks();
+    proxy_hooks = new ProxyVariableHooks{saved_hooks};
+    at::impl::SetVariableHooks(proxy_hooks);
+  }
+}
+
+void removeProxyHooks() {
+  std::lock_guard<std::mutex> lock{hooks_mutex};
+  TORCH_INTERNAL_ASSERT(hooks_refcount > 0);
+  if (--hooks_refcount == 0) {
+    at::impl::SetVariableHooks(saved_hooks);
+    delete proxy_hooks;
+    proxy_hooks = nullptr;
+    saved_hooks = nullptr;
+  }
+}
+
+at::TensorBase ProxyVariableHooks::variable_data(
+    const at::TensorBase& self) const {
+  at::TensorBase out_base = inner_->variable_data(self);
+  at::Tensor self_t{self};
+  if (!shouldRecordHere() || !isFake(self_t)) {
+    return out_base;
+  }
+  auto* self_fake = asFake(self_t);
+  if (getRecord(self_fake) == nullptr) {
+    return out_base;
+  }
+  at::Tensor out{out_base};
+  torch::jit::Stack rets{c10::IValue{out}};
+  std::vector<c10::IValue> args{c10::IValue{self_t}};
+  recordOp(
+      "tdx::variable_data",
+      [](torch::jit::Stack& s) {
+        at::Tensor in = s.back().toTensor();
+        s.pop_back();
+        s.emplace_back(in.variable_data());
+      },
+      std::move(args), at::ThreadLocalState{}, rets, 0);
+  return out;
+}
+
+void ProxyVariableHooks::set_data(const at::TensorBase& self,
+                                  const at::TensorBase& new_data) const {
+  at::Tensor self_t{self};
+  at::Tensor new_t{new_data};
+  bool record = shouldRecordHere() && isFake(self_t) &&
+                getRecord(asFake(self_t)) != nullptr;
+  std::vector<c10::IValue> args;
+  if (record) {
+    // Capture the pre-mutation producer of `self` before the impl swap.
+    args = {c10::IValue{self_t}, c10::IValue{new_t}};
+  }
+  inner_->set_data(self, new_data);
+  if (!record) {
+    return;
+  }
+  torch::jit::Stack rets{c10::IValue{self_t}};
+  recordOp(
+      "tdx::set_data",
+      [](torch::jit::Stack& s) {
+        at::Tensor nd = s.back().toTensor();
+        s.pop_back();
+        at::Tensor slf = s.back().toTensor();
+        s.pop_back();
+        slf.set_data(nd);
+        s.emplace_back(slf);
+      },
+      std::move(args), at::ThreadLocalState{}, rets, 0);
+}
+
+// ---------------------------------------------------------------------------
+// Materialization.
+// ---------------------------------------------------------------------------
+
+// Does `node` produce (as one of its outputs) a tensor aliasing `storage`?
+bool outputsAlias(const OpNode& node, const c10::Storage& storage) {
+  for (const c10::Storage& s : node.output_storages) {
+    if (storagesAlias(s, storage)) {
+      return true;
+    }
+  }
+  return false;
+}
+
+bool sharesOutputStorage(const OpNode& a, const OpNode& b) {
+  for (const c10::Storage& s : a.output_storages) {
+    if (s && outputsAlias(b, s)) {
+      return true;
+    }
+  }
+  return false;
+}
+
+// Chronologically last node (transitively reachable through dependents)
+// whose outputs alias `storage` (reference getLastInPlaceOpNode,
+// deferred_init.cc:541-579).
+std::shared_ptr<OpNode> findLastAliasWriter(
+    const std::shared_ptr<OpNode>& start, const c10::Storage& storage) {
+  std::shared_ptr<OpNode> last = start;
+  std::unordered_set<OpNode*> visited{start.get()};
+  std::vector<std::shared_ptr<OpNode>> frontier{start};
+  while (!frontier.empty()) {
+    auto n = std::move(frontier.back());
+    frontier.pop_back();
+    for (const auto& wd : n->dependents) {
+      auto d = wd.lock();
+      if (!d || !visited.insert(d.get()).second) {
+        continue;
+      }
+      if (d->op_nr > last->op_nr && outputsAlias(*d, storage)) {
+        last = d;
+      }
+      frontier.push_back(std::move(d));
+    }
+  }
+  return last;
+}
+
+struct CallStack {
+  std::vector<std::shared_ptr<OpNode>> nodes;
+  std::unordered_set<OpNode*> members;
+
+  void addWithDeps(const std::shared_ptr<OpNode>& n) {
+    if (!members.insert(n.get()).second) {
+      return;
+    }
+    nodes.push_back(n);
+    for (const InputSlot& slot : n->input_slots) {
+      if (slot.desc.has_value()) {
+        addWithDeps(slot.desc->node);
+      }
+    }
+  }
+};
+
+// Builds the ordered replay set for `target` (reference buildCallStack +
+// collectCallStack, deferred_init.cc:530-622): the dependency closure of
+// the target and of the last writer aliasing the target's storage, plus —
+// to fixpoint — (a) dependents up to that writer that share storage with a
+// collected node (in-place writers and views of the same storage) and
+// (b) dependents whose inputs a later collected writer would clobber,
+// which therefore must replay now or never see the right values.
+std::vector<std::shared_ptr<OpNode>> buildCallStack(
+    const std::shared_ptr<OpNode>& target, const c10::Storage& storage) {
+  auto last = findLastAliasWriter(target, storage);
+
+  CallStack cs;
+  cs.addWithDeps(target);
+  cs.addWithDeps(last);
+
+  bool changed = true;
+  while (changed) {
+    changed = false;
+    // Snapshot: addWithDeps mutates cs.nodes.
+    std::vector<std::shared_ptr<OpNode>> snapshot = cs.nodes;
+    for (const auto& n : snapshot) {
+      for (const auto& wd : n->dependents) {
+        auto d = wd.lock();
+        if (!d || cs.members.count(d.get()) != 0 || d->materialized) {
+          continue;
+        }
+        bool include = false;
+        if (d->op_nr <= last->op_nr && sharesOutputStorage(*d, *n)) {
+          include = true;
+        } else {
+          // Clobber rule: `d` reads an output of a collected node whose
+          // storage a collected later op overwrites.
+          for (const InputSlot& slot : d->input_slots) {
+            if (!slot.desc.has_value() ||
+                cs.members.count(slot.desc->node.get()) == 0) {
+              continue;
+            }
+            const c10::Storage& s =
+                slot.desc->node->output_storages[slot.desc->index];
+            if (!s) {
+              continue;
+            }
+            for (const auto& w : cs.nodes) {
+              if (w->op_nr > d->op_nr && w.get() != slot.desc->node.get() &&
+                  outputsAlias(*w, s)) {
+                include = true;
+                break;
+              }
+            }
+            if (include) {
+              break;
+            }
+          }
+        }
+        if (include) {
+          cs.addWithDeps(d);
+          changed = true;
+        }
+      }
+    }
+  }
+
+  std::sort(cs.nodes.begin(), cs.nodes.end(),
+            [](const auto& a, const auto& b) { return a->op_nr < b->op_nr; });
+  return cs.nodes;
+}
+
+// Rebuilds a replay stack for `node` from its saved frame, substituting
+// materialized dependency outputs for fake placeholders and validating
+// external tensors (reference materializeArguments,
+// deferred_init.cc:640-667).
+torch::jit::Stack materializeArguments(OpNode& node) {
+  torch::jit::Stack stack{node.op->args.begin(), node.op->args.end()};
+  size_t visit = 0;
+  mapTensors(stack, 0, stack.size(), [&](const at::Tensor& t) -> at::Tensor {
+    const InputSlot& slot = node.input_slots.at(visit++);
+    if (slot.desc.has_value()) {
+      const OpOutputDescriptor& dep = *slot.desc;
+      TORCH_INTERNAL_ASSERT(dep.node->materialized,
+                            "deferred-init replay ordering bug: dependency "
+                            "not yet materialized");
+      return dep.node->outputs.at(dep.index);
+    }
+    if (t.defined() && slot.external_version.has_value()) {
+      TORCH_CHECK(!t.is_inference(),
+                  "An inference tensor was used while recording `",
+                  node.op->name,
+                  "`; inference tensors cannot participate in deferred "
+                  "initialization.");
+      TORCH_CHECK(tensorVersion(t) == *slot.external_version,
+                  "The external tensor used by the recorded op `",
+                  node.op->name,
+                  "` was modified in place after recording; the tape can no "
+                  "longer be replayed faithfully.");
+    }
+    return t;
+  });
+  return stack;
+}
+
+void replayNode(const std::shared_ptr<OpNode>& node) {
+  if (node->materialized) {
+    return;
+  }
+  TORCH_CHECK(node->op.has_value(),
+              "This deferred-init tape segment has already been freed and "
+              "cannot be replayed again.");
+  torch::jit::Stack stack = materializeArguments(*node);
+  {
+    at::ThreadLocalStateGuard tls_guard{node->op->tls};
+    // Replay must run for real regardless of any ambient fake/deferred
+    // mode captured in the snapshot or active on this thread.
+    c10::impl::ExcludeDispatchKeyGuard no_deferred{kDeferredKey};
+    c10::impl::ExcludeDispatchKeyGuard no_fake{kFakeKey};
+    node->op->run(stack);
+  }
+  node->outputs.clear();
+  visitTensors(stack, 0, stack.size(),
+               [&](const at::Tensor& t) { node->outputs.push_back(t); });
+  node->materialized = true;
+  // Free the tape incrementally: drop the call frame and the dependency
+  // edges (reference detachDependencies, deferred_init.cc:524).
+  node->op.reset();
+  node->input_slots.clear();
+  node->input_slots.shrink_to_fit();
+}
+
+}  // namespace
+
+void enterDeferredInit() {
+  if (deferred_level++ == 0) {
+    c10::impl::tls_set_dispatch_key_included(kDeferredKey, true);
+    installProxyHooks();
+  }
+}
+
+void leaveDeferredInit() {
+  TORCH_CHECK(deferred_level > 0, "Not in a deferred-init context.");
+  if (--deferred_level == 0) {
+    c10::impl::tls_set_dispatch_key_included(kDeferredKey, false);
+    removeProxyHooks();
+  }
+}
+
+bool isDeferredInitActive() noexcept {
+  return deferred_level > 0;
+}
+
+NoDeferredInit::NoDeferredInit()
+    : prev_{c10::impl::tls_is_dispatch_key_excluded(kDeferredKey)} {
+  c10::impl::tls_set_dispatch_key_excluded(kDeferredKey, true);
+}
+
+NoDeferredInit::~NoDeferredInit() {
+  c10::impl::tls_set_dispatch_key_excluded(kDeferredKey, prev_);
+}
+
+bool canMaterialize(const at::Tensor& tensor) noexcept {
+  auto* fake = asFake(tensor);
+  return fake != nullptr && getRecord(fake) != nullptr;
+}
+
+at::Tensor materializeTensor(const at::Tensor& tensor) {
+  auto* fake = asFake(tensor);
+  if (fake == nullptr) {
+    return tensor;  // identity for real tensors
+  }
+  auto rec = getRecord(fake);
+  TORCH_CHECK_VALUE(rec != nullptr,
+                    "`tensor` is fake but carries no deferred-init record, "
+                    "so it cannot be materialized.");
+
+  auto node = rec->desc.node;
+  const c10::Storage& storage = node->output_storages.at(rec->desc.index);
+  for (const auto& n : buildCallStack(node, storage)) {
+    replayNode(n);
+  }
+
+  at::Tensor out = node->outputs.at(rec->desc.index);
+  // requires_grad_() is untraceable on purpose; re-apply the fake tensor's
+  // final autograd flag explicitly (reference deferred_init.cc:713-729).
+  if (out.is_leaf() && out.requires_grad() != tensor.requires_grad()) {
+    out.set_requires_grad(tensor.requires_grad());
+  }
+  return out;
+}
+
+}  // namespace tdx

@@ -1,0 +1,45 @@
+// Deferred module initialization: record every op performed on fake tensors
+// on an in-memory tape, then replay the tape to materialize real tensors.
+//
+// Capability parity with the reference deferred-init core
+// (/root/reference/src/cc/torchdistx/deferred_init.{h,cc}): a boxed fallback
+// on the pre-autograd `DeferredInit` dispatch key, an Op/OpNode tape with
+// chronological (op_nr) ordering, storage-alias-aware in-place replay,
+// view keep-alives, external-tensor version-counter checks, a
+// VariableHooks proxy recording `.data` reads/writes, and identity-stable
+// repeated materialization.
+
+#pragma once
+
+#include <ATen/ATen.h>
+
+namespace tdx {
+
+// TLS-scoped deferred-init mode; nestable.
+void enterDeferredInit();
+void leaveDeferredInit();
+bool isDeferredInitActive() noexcept;
+
+// RAII: temporarily disable deferred-init recording on this thread.
+class NoDeferredInit {
+ public:
+  NoDeferredInit();
+  ~NoDeferredInit();
+  NoDeferredInit(const NoDeferredInit&) = delete;
+  NoDeferredInit& operator=(const NoDeferredInit&) = delete;
+
+ private:
+  bool prev_;
+};
+
+// True when `tensor` is a fake tensor carrying a deferred-init record (i.e.
+// materializeTensor can reconstruct it).
+bool canMaterialize(const at::Tensor& tensor) noexcept;
+
+// Replays the relevant tape segment and returns the real tensor. Identity
+// rules: a non-fake input is returned as-is; repeated calls (and aliased
+// fakes) yield the same underlying TensorImpl. Raises c10::ValueError for a
+// fake tensor with no record.
+at::Tensor materializeTensor(const at::Tensor& tensor);
+
+}  // namespace tdx

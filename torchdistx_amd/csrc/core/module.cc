@@ -1,0 +1,84 @@
+// Python bindings for the fake-tensor and deferred-init cores.
+//
+// Capability parity with the reference binding layer
+// (/root/reference/src/python/torchdistx/_C/{module,fake,deferred_init}.cc):
+// the same eight entry points (enter/leave_deferred_init,
+// enter/leave_fake_mode, is_fake, can_materialize, materialize_tensor,
+// meta_like), GIL release around materialization, ValueError translation,
+// and CUDA lazy-init suppression for the fake-cuda mode. Python-object
+// identity across repeated materializations comes from THPVariable_Wrap's
+// pyobj-slot reuse (one TensorImpl <-> one Python object); Parameter-class
+// preservation is handled in the Python API layer.
+
+#include <torch/extension.h>
+
+#include <torch/csrc/utils/device_lazy_init.h>
+
+#include "deferred_init.h"
+#include "fake.h"
+
+namespace {
+
+// Tracks whether we disabled CUDA lazy-init so nested fake modes restore it
+// exactly once.
+thread_local size_t fake_cuda_lazy_init_suppressions = 0;
+
+void enterFakeModeBinding(bool fake_cuda) {
+  if (fake_cuda) {
+    // Keep tensor construction from eagerly initializing the (possibly
+    // absent) HIP runtime while fake "cuda" tensors are being built.
+    torch::utils::set_requires_device_init(at::kCUDA, false);
+    ++fake_cuda_lazy_init_suppressions;
+  }
+  tdx::enterFakeMode(fake_cuda);
+}
+
+void leaveFakeModeBinding() {
+  tdx::leaveFakeMode();
+  if (!tdx::isFakeModeActive() && fake_cuda_lazy_init_suppressions > 0) {
+    fake_cuda_lazy_init_suppressions = 0;
+    torch::utils::set_requires_device_init(at::kCUDA, true);
+  }
+}
+
+}  // namespace
+
+PYBIND11_MODULE(_C, m) {
+  m.doc() = "torchdistx_amd native core: fake tensors + deferred init";
+
+  pybind11::register_exception_translator([](std::exception_ptr e) {
+    try {
+      if (e) {
+        std::rethrow_exception(e);
+      }
+    } catch (const c10::ValueError& err) {
+      PyErr_SetString(PyExc_ValueError, err.what_without_backtrace());
+    }
+    // Other exception types fall through to the previously registered
+    // translators (pybind runs translators in reverse registration order).
+  });
+
+  m.def("enter_fake_mode", &enterFakeModeBinding, pybind11::arg("fake_cuda"));
+  m.def("leave_fake_mode", &leaveFakeModeBinding);
+
+  m.def("is_fake",
+        [](const at::Tensor& t) { return tdx::isFake(t); });
+
+  m.def("meta_like",
+        [](const at::Tensor& t) { return tdx::metaLike(t); });
+
+  m.def("enter_deferred_init", &tdx::enterDeferredInit);
+  m.def("leave_deferred_init", &tdx::leaveDeferredInit);
+
+  m.def("can_materialize",
+        [](const at::Tensor& t) { return tdx::canMaterialize(t); });
+
+  m.def("materialize_tensor", [](const at::Tensor& t) {
+    at::Tensor out;
+    {
+      pybind11::gil_scoped_release release;
+      out = tdx::materializeTensor(t);
+    }
+    return out;
+  });
+}

@@ -1,0 +1,134 @@
+# Deferred-initialization Python API.
+#
+# Capability parity with the reference
+# (/root/reference/src/python/torchdistx/deferred_init.py:19-124):
+# deferred_init / is_deferred / materialize_tensor / materialize_module with
+# identical semantics, including recursive child-first module
+# materialization, buffers_only / check_fn filtering, object identity for
+# repeated and aliased materializations, and in-place swapping of
+# module._parameters / module._buffers entries.
+#
+# Design difference vs the reference binding layer: the reference allocates
+# the materialized Python object with the original's class in C++
+# (_C/deferred_init.cc:33-94); here the C++ core returns the canonical
+# Tensor wrapper (stable per TensorImpl) and the Parameter class is restored
+# in Python via torch.nn.Parameter, memoized per materialized tensor so
+# aliased parameters still materialize to one object.
+
+from typing import Callable, Dict, Optional, TypeVar, Union
+
+import torch
+from torch import Tensor
+from torch.nn import Module, Parameter
+from torch.utils.weak import WeakTensorKeyDictionary
+
+# Imported for its Tensor.__repr__ patch.
+from torchdistx_amd import fake  # noqa: F401
+from torchdistx_amd import _C
+
+T = TypeVar("T", bound=Module)
+
+# materialized base tensor -> Parameter wrapper, so aliased parameters (and
+# repeated materializations) map to a single Parameter object.
+_parameter_memo: "WeakTensorKeyDictionary" = WeakTensorKeyDictionary()
+
+
+def deferred_init(module_fn: Callable[..., T], *args, **kwargs) -> T:
+    """Runs ``module_fn`` with tensor construction deferred: every tensor it
+    creates is fake, and every operation on those tensors is recorded on an
+    in-memory tape. The result can later be turned into a real module with
+    :func:`materialize_module` (or tensor-by-tensor with
+    :func:`materialize_tensor`) — including on a different device, sharded
+    across ranks, or through the CDNA4 HIP init kernels on an MI355X.
+
+    Args:
+        module_fn: a callable returning a ``Module``.
+        args, kwargs: forwarded to ``module_fn``.
+
+    .. warning::
+        Only operations performed *inside* ``deferred_init()`` are recorded.
+        Mutating the returned module afterwards (outside another deferred
+        context) cannot be replayed and will make materialization
+        incorrect or impossible.
+    """
+    _C.enter_deferred_init()
+    try:
+        return module_fn(*args, **kwargs)
+    finally:
+        _C.leave_deferred_init()
+
+
+def is_deferred(obj: Union[Tensor, Module]) -> bool:
+    """Whether ``obj`` (a tensor, or any parameter/buffer of a module) still
+    awaits materialization."""
+    if isinstance(obj, Tensor):
+        return _C.can_materialize(obj)
+
+    if isinstance(obj, Module):
+        return any(
+            _C.can_materialize(t)
+            for t in list(obj.parameters()) + list(obj.buffers())
+        )
+
+    raise ValueError("`obj` must be of type `Tensor` or `Module`.")
+
+
+def materialize_tensor(tensor: Tensor) -> Tensor:
+    """Materializes ``tensor`` by replaying the relevant part of its
+    recording tape. Real tensors pass through unchanged; repeated calls and
+    aliased fakes return the same object.
+
+    .. warning::
+        A materialized fake keeps a reference to its materialized value;
+        drop the fake once it is no longer needed to release memory.
+    """
+    materialized = _C.materialize_tensor(tensor)
+    if materialized is tensor:
+        return tensor
+    return _restore_class(tensor, materialized)
+
+
+def _restore_class(original: Tensor, materialized: Tensor) -> Tensor:
+    if not isinstance(original, Parameter):
+        return materialized
+    wrapper = _parameter_memo.get(materialized)
+    if wrapper is None:
+        wrapper = Parameter(materialized, requires_grad=original.requires_grad)
+        _parameter_memo[materialized] = wrapper
+    return wrapper
+
+
+def materialize_module(
+    module: Module,
+    buffers_only: bool = False,
+    check_fn: Optional[Callable[[Module], bool]] = None,
+) -> None:
+    """Materializes ``module`` in place: children first, then this module's
+    parameters and buffers.
+
+    Args:
+        module: the deferred module to materialize.
+        buffers_only: only materialize buffers.
+        check_fn: optional per-module predicate; modules for which it
+            returns ``False`` are skipped (their children are still
+            visited).
+    """
+
+    def swap(tensors: Dict[str, Optional[Tensor]]) -> None:
+        for key, tensor in tensors.items():
+            if tensor is None:
+                continue
+            try:
+                tensors[key] = materialize_tensor(tensor)
+            except ValueError:
+                raise ValueError(
+                    f"'{key}' has already been materialized."
+                ) from None
+
+    for child in module.children():
+        materialize_module(child, buffers_only, check_fn)
+
+    if check_fn is None or check_fn(module):
+        if not buffers_only:
+            swap(module._parameters)  # type: ignore[arg-type]
+        swap(module._buffers)  # type: ignore[arg-type]

@@ -1,0 +1,2 @@
+from torchdistx_amd.slowmo.slowmo_comm import SlowMoState, slowmo_hook  # noqa: F401
+from torchdistx_amd.slowmo.slowmo_optimizer import SlowMomentumOptimizer  # noqa: F401
