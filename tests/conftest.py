@@ -1,0 +1,7 @@
+import pytest  # noqa: F401
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: test requires an MI355X (or other ROCm) GPU"
+    )

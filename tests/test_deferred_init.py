@@ -1,0 +1,267 @@
+# Deferred-init tests. Coverage model: reference
+# tests/python/test_deferred_init.py (identity/no-op/lifecycle) plus the
+# in-place/view replay example from the reference docs
+# (fake_tensor_and_deferred_init.rst:197-209), RNG fidelity, .data
+# recording, external-tensor version tracking, and materialize_module
+# options.
+
+from typing import cast
+
+import pytest
+import torch
+from torch import Tensor
+from torch.nn import Module, Parameter
+
+from torchdistx_amd.deferred_init import (
+    deferred_init,
+    is_deferred,
+    materialize_module,
+    materialize_tensor,
+)
+from torchdistx_amd.fake import fake_mode, is_fake
+
+
+def test_materialize_tensor_is_noop_for_real_tensors() -> None:
+    a = torch.ones([10])
+    assert materialize_tensor(a) is a
+
+
+def test_materialize_tensor_returns_same_tensor() -> None:
+    class FooModule(Module):
+        def __init__(self):
+            super().__init__()
+            self.param1 = Parameter(torch.ones([5]))
+            self.param2 = self.param1
+
+    module = deferred_init(FooModule)
+
+    a = materialize_tensor(cast(Tensor, module.param1))
+    b = materialize_tensor(cast(Tensor, module.param1))
+    c = materialize_tensor(cast(Tensor, module.param2))
+
+    assert a is b
+    assert a is c
+    assert isinstance(a, Parameter)
+    assert torch.equal(a.detach(), torch.ones([5]))
+
+
+def test_is_deferred_returns_right_value() -> None:
+    class FooModule(Module):
+        def __init__(self):
+            super().__init__()
+            self.param1 = Parameter(torch.ones([5]))
+            self.param2 = Parameter(torch.ones([5]))
+
+    module = FooModule()
+    assert not is_deferred(module)
+
+    module = deferred_init(FooModule)
+    assert is_deferred(module)
+
+    materialize_module(module)
+    assert not is_deferred(module)
+
+    module = deferred_init(FooModule)
+    module.param1 = materialize_tensor(module.param1)
+    assert is_deferred(module)
+
+    module.param2 = materialize_tensor(module.param2)
+    assert not is_deferred(module)
+
+
+def test_is_deferred_raises_on_wrong_type() -> None:
+    with pytest.raises(ValueError):
+        is_deferred("not a tensor")  # type: ignore[arg-type]
+
+
+def test_materialized_module_matches_eager_init() -> None:
+    torch.manual_seed(1234)
+    deferred = deferred_init(torch.nn.Linear, 16, 32)
+    materialize_module(deferred)
+
+    torch.manual_seed(1234)
+    eager = torch.nn.Linear(16, 32)
+
+    assert torch.equal(deferred.weight, eager.weight)
+    assert torch.equal(deferred.bias, eager.bias)
+    assert deferred.weight.requires_grad
+    assert isinstance(deferred.weight, Parameter)
+
+
+def test_materialized_sequential_matches_eager_init() -> None:
+    def build():
+        return torch.nn.Sequential(
+            torch.nn.Linear(8, 8),
+            torch.nn.LayerNorm(8),
+            torch.nn.Linear(8, 4),
+        )
+
+    torch.manual_seed(99)
+    deferred = deferred_init(build)
+    assert is_deferred(deferred)
+    materialize_module(deferred)
+
+    torch.manual_seed(99)
+    eager = build()
+
+    for (dn, dp), (en, ep) in zip(
+        deferred.named_parameters(), eager.named_parameters()
+    ):
+        assert dn == en
+        assert torch.equal(dp, ep), dn
+
+
+def test_inplace_view_replay() -> None:
+    # The docs' canonical aliasing example: a view must observe in-place
+    # updates to its base that happened after the view was created.
+    class M(Module):
+        def __init__(self):
+            super().__init__()
+            a = torch.zeros([4])
+            v = a.view(2, 2)
+            a.add_(1)
+            self.p = Parameter(v)
+
+    m = deferred_init(M)
+    p = materialize_tensor(cast(Tensor, m.p))
+    assert torch.equal(p.detach(), torch.ones([2, 2]))
+
+
+def test_inplace_after_view_on_view_replay() -> None:
+    class M(Module):
+        def __init__(self):
+            super().__init__()
+            a = torch.zeros([4])
+            v = a.view(2, 2)
+            v.fill_(3)
+            self.base = Parameter(a)
+
+    m = deferred_init(M)
+    base = materialize_tensor(cast(Tensor, m.base))
+    assert torch.equal(base.detach(), torch.full([4], 3.0))
+
+
+def test_dot_data_assignment_is_replayed() -> None:
+    class M(Module):
+        def __init__(self):
+            super().__init__()
+            self.p = Parameter(torch.empty(3))
+            self.p.data.fill_(2.0)
+            self.p.data = torch.full([3], 7.0)
+
+    m = deferred_init(M)
+    p = materialize_tensor(cast(Tensor, m.p))
+    assert torch.equal(p.detach(), torch.full([3], 7.0))
+
+
+def test_item_is_terminal() -> None:
+    class M(Module):
+        def __init__(self):
+            super().__init__()
+            scale = (torch.ones(1) * 0.5).item()
+            self.p = Parameter(torch.full([2], scale))
+
+    m = deferred_init(M)
+    p = materialize_tensor(cast(Tensor, m.p))
+    assert torch.equal(p.detach(), torch.full([2], 0.5))
+
+
+def test_external_cpu_scalar_version_check() -> None:
+    ext = torch.tensor(2.0)
+
+    class M(Module):
+        def __init__(self):
+            super().__init__()
+            self.p = Parameter(torch.zeros([3]) + ext)
+
+    good = deferred_init(M)
+    assert torch.equal(
+        materialize_tensor(cast(Tensor, good.p)).detach(), torch.full([3], 2.0)
+    )
+
+    stale = deferred_init(M)
+    ext.add_(1)
+    with pytest.raises(RuntimeError, match="modified in place"):
+        materialize_tensor(cast(Tensor, stale.p))
+
+
+def test_materialize_module_buffers_only() -> None:
+    class M(Module):
+        def __init__(self):
+            super().__init__()
+            self.p = Parameter(torch.ones([2]))
+            self.register_buffer("b", torch.zeros([2]))
+
+    m = deferred_init(M)
+    materialize_module(m, buffers_only=True)
+    assert is_fake(m.p)
+    assert not is_fake(m.b)
+    materialize_module(m)
+    assert not is_deferred(m)
+
+
+def test_materialize_module_check_fn() -> None:
+    class Leaf(Module):
+        def __init__(self):
+            super().__init__()
+            self.p = Parameter(torch.ones([2]))
+
+    class Root(Module):
+        def __init__(self):
+            super().__init__()
+            self.skip = Leaf()
+            self.keep = Leaf()
+
+    m = deferred_init(Root)
+    materialize_module(m, check_fn=lambda mod: mod is not m.skip)
+    assert is_fake(m.skip.p)
+    assert not is_fake(m.keep.p)
+
+
+def test_requires_grad_is_preserved() -> None:
+    class M(Module):
+        def __init__(self):
+            super().__init__()
+            self.p = Parameter(torch.ones([2]), requires_grad=False)
+            self.register_buffer("b", torch.zeros([2]))
+
+    m = deferred_init(M)
+    materialize_module(m)
+    assert not m.p.requires_grad
+    assert not m.b.requires_grad
+
+
+def test_fake_without_record_raises_value_error() -> None:
+    with fake_mode():
+        a = torch.ones([3])
+    with pytest.raises(ValueError):
+        materialize_tensor(a)
+
+
+def test_fake_from_plain_fake_mode_rejected_in_deferred_init() -> None:
+    with fake_mode():
+        orphan = torch.ones([3])
+
+    def build():
+        class M(Module):
+            def __init__(self):
+                super().__init__()
+                self.p = Parameter(orphan + 1)
+
+        return M()
+
+    with pytest.raises(RuntimeError, match="deferred-init"):
+        deferred_init(build)
+
+
+def test_deferred_init_forward_shapes_on_fake() -> None:
+    # A forward pass on a fully fake module stays fake and shape-correct.
+    def build():
+        return torch.nn.Linear(6, 3)
+
+    m = deferred_init(build)
+    with fake_mode():
+        x = torch.randn(5, 6)
+        y = m(x)
+    assert is_fake(y)
+    assert y.shape == (5, 3)
