@@ -19,3 +19,7 @@ from torchdistx_amd.deferred_init import (  # noqa: F401
     materialize_module,
     materialize_tensor,
 )
+
+# Register the CDNA4 kernels (tdx:: ops) when the extension is built; the
+# deferred-init replay engine redirects GPU init ops to them.
+from torchdistx_amd import _kernels  # noqa: E402,F401

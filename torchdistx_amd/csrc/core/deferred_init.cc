@@ -31,6 +31,7 @@
 #include <torch/library.h>
 
 #include "fake.h"
+#include "native_redirect.h"
 #include "stack_utils.h"
 
 namespace tdx {
@@ -250,7 +251,15 @@ void deferredInitHandler(const c10::OperatorHandle& op,
   if (has_fake_arg || has_fake_ret) {
     recordOp(
         schema.operator_name().name,
-        [handle = op](torch::jit::Stack& s) { handle.callBoxed(s); },
+        [handle = op](torch::jit::Stack& s) {
+          // Replay hot path: recorded init ops whose target lives on the
+          // GPU run through the hand-written CDNA4 kernels (tdx::) when
+          // the _K extension is loaded.
+          if (tryNativeInitRedirect(handle, s)) {
+            return;
+          }
+          handle.callBoxed(s);
+        },
         std::move(saved), std::move(tls), *stack, rets_begin);
   }
 }

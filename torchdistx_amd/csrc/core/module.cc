@@ -16,6 +16,7 @@
 
 #include "deferred_init.h"
 #include "fake.h"
+#include "native_redirect.h"
 
 namespace {
 
@@ -72,6 +73,10 @@ PYBIND11_MODULE(_C, m) {
 
   m.def("can_materialize",
         [](const at::Tensor& t) { return tdx::canMaterialize(t); });
+
+  m.def("set_native_init", &tdx::setNativeInitEnabled,
+        pybind11::arg("enabled"));
+  m.def("native_init_enabled", &tdx::nativeInitEnabled);
 
   m.def("materialize_tensor", [](const at::Tensor& t) {
     at::Tensor out;

@@ -1,0 +1,88 @@
+#include "native_redirect.h"
+
+#include <atomic>
+#include <cstdlib>
+#include <cstring>
+
+namespace tdx {
+namespace {
+
+std::atomic<bool> native_init_enabled{true};
+
+struct Redirect {
+  const char* aten_name;
+  const char* aten_overload;
+  const char* tdx_name;
+  bool needs_float_dtype;  // RNG/fill kernels cover f32/bf16/f16 only
+};
+
+constexpr Redirect kRedirects[] = {
+    {"aten::uniform_", "", "uniform_", true},
+    {"aten::normal_", "", "normal_", true},
+    {"aten::fill_", "Scalar", "fill_", true},
+    {"aten::zero_", "", "zero_", false},
+};
+
+bool requireNative() {
+  static const bool required = std::getenv("TDX_REQUIRE_NATIVE_INIT") != nullptr;
+  return required;
+}
+
+}  // namespace
+
+void setNativeInitEnabled(bool enabled) noexcept {
+  native_init_enabled.store(enabled, std::memory_order_relaxed);
+}
+
+bool nativeInitEnabled() noexcept {
+  return native_init_enabled.load(std::memory_order_relaxed);
+}
+
+bool tryNativeInitRedirect(const c10::OperatorHandle& op,
+                           torch::jit::Stack& stack) {
+  if (!nativeInitEnabled()) {
+    return false;
+  }
+  const auto& name = op.schema().operator_name();
+  const Redirect* redirect = nullptr;
+  for (const Redirect& r : kRedirects) {
+    if (name.name == r.aten_name && name.overload_name == r.aten_overload) {
+      redirect = &r;
+      break;
+    }
+  }
+  if (redirect == nullptr) {
+    return false;
+  }
+
+  // These schemas all take `self` as the first (and mutated) argument; the
+  // replay stack holds exactly the op's arguments at this point.
+  if (stack.empty() || !stack.front().isTensor()) {
+    return false;
+  }
+  const at::Tensor& self = stack.front().toTensor();
+  if (!self.defined() || !self.is_cuda() || !self.is_contiguous()) {
+    return false;
+  }
+  if (redirect->needs_float_dtype) {
+    auto st = self.scalar_type();
+    if (st != at::kFloat && st != at::kBFloat16 && st != at::kHalf) {
+      return false;
+    }
+  }
+
+  auto target = c10::Dispatcher::singleton().findOp(
+      {std::string("tdx::") + redirect->tdx_name, ""});
+  if (!target.has_value()) {
+    TORCH_CHECK(!requireNative(),
+                "TDX_REQUIRE_NATIVE_INIT is set but the torchdistx_amd._K "
+                "kernel extension is not loaded; refusing to fall back to "
+                "stock ATen kernels for `",
+                name, "`.");
+    return false;
+  }
+  target->callBoxed(&stack);
+  return true;
+}
+
+}  // namespace tdx

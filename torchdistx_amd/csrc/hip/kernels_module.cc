@@ -1,0 +1,48 @@
+// Python module for the CDNA4 kernel extension (torchdistx_amd._K).
+// Importing this module also registers the tdx:: init ops with the
+// dispatcher (TORCH_LIBRARY in init_kernels.hip), which is what the
+// deferred-init replay redirect looks for.
+
+#include <torch/extension.h>
+
+namespace tdx {
+
+void anyprecision_adamw_step(at::Tensor& param,
+                             const at::Tensor& grad,
+                             at::Tensor& exp_avg,
+                             at::Tensor& exp_avg_sq,
+                             std::optional<at::Tensor> compensation,
+                             double lr,
+                             double beta1,
+                             double beta2,
+                             double eps,
+                             double weight_decay,
+                             double step_size,
+                             double bias_correction2_sqrt);
+
+}  // namespace tdx
+
+PYBIND11_MODULE(_K, m) {
+  m.doc() = "torchdistx_amd CDNA4 kernels (gfx950)";
+
+  m.def("has_init_kernels", [] { return true; });
+  m.def("has_anyprecision_adamw", [] { return true; });
+
+  m.def(
+      "anyprecision_adamw_",
+      [](at::Tensor param, at::Tensor grad, at::Tensor exp_avg,
+         at::Tensor exp_avg_sq, std::optional<at::Tensor> compensation,
+         double lr, double beta1, double beta2, double eps,
+         double weight_decay, double step_size,
+         double bias_correction2_sqrt) {
+        tdx::anyprecision_adamw_step(
+            param, grad, exp_avg, exp_avg_sq, std::move(compensation), lr,
+            beta1, beta2, eps, weight_decay, step_size,
+            bias_correction2_sqrt);
+      },
+      pybind11::arg("param"), pybind11::arg("grad"), pybind11::arg("exp_avg"),
+      pybind11::arg("exp_avg_sq"), pybind11::arg("compensation"),
+      pybind11::arg("lr"), pybind11::arg("beta1"), pybind11::arg("beta2"),
+      pybind11::arg("eps"), pybind11::arg("weight_decay"),
+      pybind11::arg("step_size"), pybind11::arg("bias_correction2_sqrt"));
+}
