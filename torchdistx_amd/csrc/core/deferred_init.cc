@@ -24,9 +24,11 @@
 #include <unordered_set>
 #include <vector>
 
+#include <ATen/Context.h>
 #include <ATen/ThreadLocalState.h>
 #include <ATen/core/dispatch/Dispatcher.h>
 #include <ATen/core/VariableHooksInterface.h>
+#include <c10/core/DefaultDtype.h>
 #include <c10/core/impl/LocalDispatchKeySet.h>
 #include <torch/library.h>
 
@@ -65,6 +67,32 @@ struct RecordedOp {
   std::vector<c10::IValue> args;  // deep-copied; fake slots -> undefined
   at::ThreadLocalState tls;       // captured with DeferredInit excluded
 };
+
+// Pinned Philox state for recorded RNG ops: derived from the default
+// generator's seed and the op's tape position at RECORD time, so replaying
+// ANY subset of the tape on ANY rank produces identical bits
+// (partition-invariant initialization — the property that makes sharded /
+// broadcast materialization bitwise-equal to local materialization when
+// the native CDNA4 kernels run the replay).
+constexpr uint64_t kPhiloxStridePerOp = 4;
+
+std::atomic<uint64_t> next_rng_slot{1};
+
+bool isRngOpName(const std::string& name) {
+  return name == "aten::uniform_" || name == "aten::normal_";
+}
+
+std::optional<std::pair<uint64_t, uint64_t>> pinPhiloxForOp(
+    const std::string& name) {
+  if (!isRngOpName(name)) {
+    return std::nullopt;
+  }
+  uint64_t seed = at::globalContext()
+                      .defaultGenerator(c10::DeviceType::CPU)
+                      .current_seed();
+  uint64_t slot = next_rng_slot.fetch_add(1, std::memory_order_relaxed);
+  return std::make_pair(seed, slot * kPhiloxStridePerOp);
+}
 
 struct InputSlot {
   std::optional<OpOutputDescriptor> desc;  // set when the arg was fake
@@ -238,6 +266,32 @@ void deferredInitHandler(const c10::OperatorHandle& op,
   auto saved = copyStackRegion(*stack, args_begin, args_begin + num_args);
   at::ThreadLocalState tls;  // DeferredInit is excluded in this snapshot
 
+  // Factory calls (no tensor inputs) with dtype=None resolve the dtype at
+  // execution time from the process-global default. Pin the default that
+  // is in effect NOW into the recorded frame, so replay is faithful even
+  // after torch.set_default_dtype() changes (for non-factories None means
+  // "follow the input tensor" and must stay None).
+  bool has_tensor_arg = false;
+  visitTensors(*stack, args_begin, args_begin + num_args,
+               [&](const at::Tensor& t) {
+                 has_tensor_arg = has_tensor_arg || t.defined();
+               });
+  if (!has_tensor_arg) {
+    const auto& schema_args = schema.arguments();
+    for (size_t i = 0; i < schema_args.size(); ++i) {
+      if (schema_args[i].name() == "dtype" && saved[i].isNone()) {
+        // NB: schema .type() erases ScalarType to int; the semantic type
+        // lives in .real_type().
+        const auto* type = schema_args[i].real_type().get();
+        if (type->kind() == c10::TypeKind::OptionalType &&
+            type->castRaw<c10::OptionalType>()->getElementType()->kind() ==
+                c10::TypeKind::ScalarTypeType) {
+          saved[i] = c10::IValue{c10::get_default_dtype_as_scalartype()};
+        }
+      }
+    }
+  }
+
   // Execute through the fake layer: shape/dtype work happens on the meta
   // backend and fake tensors come back.
   op.redispatchBoxed(below.add(kFakeKey), stack);
@@ -249,13 +303,15 @@ void deferredInitHandler(const c10::OperatorHandle& op,
   });
 
   if (has_fake_arg || has_fake_ret) {
+    auto philox = pinPhiloxForOp(schema.operator_name().name);
     recordOp(
         schema.operator_name().name,
-        [handle = op](torch::jit::Stack& s) {
+        [handle = op, philox](torch::jit::Stack& s) {
           // Replay hot path: recorded init ops whose target lives on the
           // GPU run through the hand-written CDNA4 kernels (tdx::) when
-          // the _K extension is loaded.
-          if (tryNativeInitRedirect(handle, s)) {
+          // the _K extension is loaded. The pinned Philox state makes the
+          // result independent of which rank replays which subset.
+          if (tryNativeInitRedirect(handle, s, philox)) {
             return;
           }
           handle.callBoxed(s);

@@ -38,8 +38,10 @@ bool nativeInitEnabled() noexcept {
   return native_init_enabled.load(std::memory_order_relaxed);
 }
 
-bool tryNativeInitRedirect(const c10::OperatorHandle& op,
-                           torch::jit::Stack& stack) {
+bool tryNativeInitRedirect(
+    const c10::OperatorHandle& op,
+    torch::jit::Stack& stack,
+    const std::optional<std::pair<uint64_t, uint64_t>>& philox) {
   if (!nativeInitEnabled()) {
     return false;
   }
@@ -80,6 +82,19 @@ bool tryNativeInitRedirect(const c10::OperatorHandle& op,
                 "stock ATen kernels for `",
                 name, "`.");
     return false;
+  }
+  const bool is_rng = std::strcmp(redirect->tdx_name, "uniform_") == 0 ||
+                      std::strcmp(redirect->tdx_name, "normal_") == 0;
+  if (is_rng) {
+    // tdx RNG schemas carry two trailing optional args: the Philox seed
+    // and counter offset pinned at record time (partition-invariant init).
+    if (philox.has_value()) {
+      stack.emplace_back(static_cast<int64_t>(philox->first));
+      stack.emplace_back(static_cast<int64_t>(philox->second));
+    } else {
+      stack.emplace_back();
+      stack.emplace_back();
+    }
   }
   target->callBoxed(&stack);
   return true;

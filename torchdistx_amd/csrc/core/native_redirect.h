@@ -12,6 +12,10 @@
 
 #pragma once
 
+#include <cstdint>
+#include <optional>
+#include <utility>
+
 #include <ATen/core/dispatch/Dispatcher.h>
 #include <ATen/core/stack.h>
 
@@ -23,8 +27,10 @@ namespace tdx {
 // TDX_REQUIRE_NATIVE_INIT environment variable is set, the op is
 // redirectable, the tensor is on the GPU, and the kernel extension is not
 // loaded (fail-loud mode for GPU CI).
-bool tryNativeInitRedirect(const c10::OperatorHandle& op,
-                           torch::jit::Stack& stack);
+bool tryNativeInitRedirect(
+    const c10::OperatorHandle& op,
+    torch::jit::Stack& stack,
+    const std::optional<std::pair<uint64_t, uint64_t>>& philox = std::nullopt);
 
 void setNativeInitEnabled(bool enabled) noexcept;
 bool nativeInitEnabled() noexcept;

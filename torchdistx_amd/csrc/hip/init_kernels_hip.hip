@@ -234,7 +234,9 @@ template <Dist kDist>
 void launchRng(at::Tensor& self,
                double p0,
                double p1,
-               const std::optional<at::Generator>& generator) {
+               const std::optional<at::Generator>& generator,
+               std::optional<int64_t> pinned_seed,
+               std::optional<int64_t> pinned_offset) {
   TORCH_CHECK(self.is_contiguous(),
               "tdx init kernels require contiguous tensors");
   const int64_t n = self.numel();
@@ -242,7 +244,15 @@ void launchRng(at::Tensor& self,
     return;
   }
   uint64_t seed = 0;
-  uint64_t offset = acquireSeedOffset(generator, &seed);
+  uint64_t offset = 0;
+  if (pinned_seed.has_value() && pinned_offset.has_value()) {
+    // Replay with record-time-pinned Philox state: the result is the same
+    // no matter which rank replays this op, or in what order.
+    seed = static_cast<uint64_t>(*pinned_seed);
+    offset = static_cast<uint64_t>(*pinned_offset);
+  } else {
+    offset = acquireSeedOffset(generator, &seed);
+  }
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA();
 
   // p0/p1 are (from, to) for uniform and (mean, std) for normal; the kernel
@@ -280,19 +290,23 @@ void launchRng(at::Tensor& self,
 at::Tensor& tdx_uniform_(at::Tensor& self,
                          double from,
                          double to,
-                         std::optional<at::Generator> generator) {
+                         std::optional<at::Generator> generator,
+                         std::optional<int64_t> seed,
+                         std::optional<int64_t> offset) {
   // double (and other unsupported dtypes) never reach here: the redirect
   // layer filters, and direct callers get a clear dispatch error.
-  launchRng<Dist::kUniform>(self, from, to, generator);
+  launchRng<Dist::kUniform>(self, from, to, generator, seed, offset);
   return self;
 }
 
 at::Tensor& tdx_normal_(at::Tensor& self,
                         double mean,
                         double std,
-                        std::optional<at::Generator> generator) {
+                        std::optional<at::Generator> generator,
+                        std::optional<int64_t> seed,
+                        std::optional<int64_t> offset) {
   TORCH_CHECK(std >= 0.0, "normal_ expects std >= 0.0, but found std=", std);
-  launchRng<Dist::kNormal>(self, mean, std, generator);
+  launchRng<Dist::kNormal>(self, mean, std, generator, seed, offset);
   return self;
 }
 
@@ -346,10 +360,12 @@ at::Tensor& tdx_zero_(at::Tensor& self) {
 TORCH_LIBRARY(tdx, m) {
   m.def(
       "uniform_(Tensor(a!) self, float from=0., float to=1., *, "
-      "Generator? generator=None) -> Tensor(a!)");
+      "Generator? generator=None, int? seed=None, int? offset=None) "
+      "-> Tensor(a!)");
   m.def(
       "normal_(Tensor(a!) self, float mean=0., float std=1., *, "
-      "Generator? generator=None) -> Tensor(a!)");
+      "Generator? generator=None, int? seed=None, int? offset=None) "
+      "-> Tensor(a!)");
   m.def("fill_(Tensor(a!) self, Scalar value) -> Tensor(a!)");
   m.def("zero_(Tensor(a!) self) -> Tensor(a!)");
 }

@@ -1,0 +1,4 @@
+from torchdistx_amd.parallel.sharded_materialize import (  # noqa: F401
+    assign_owners,
+    materialize_module_distributed,
+)

@@ -1,0 +1,132 @@
+# Distributed materialization of deferred modules over RCCL/xGMI.
+#
+# This is the capability BASELINE.json adds on top of the reference (which
+# only materializes process-locally): turning one recorded deferred-init
+# tape into initialized parameters across the 8 GPUs of an MI355X node.
+#
+# Three modes, chosen for the xGMI fabric (7 point-to-point links of
+# ~153 GB/s per GPU vs ~6 TB/s of local HBM write bandwidth):
+#
+#   * "replicate" — every rank replays the full tape locally. Because the
+#     tape replays the same Philox streams from the same seed, all ranks
+#     produce bitwise-identical replicas with ZERO communication. On this
+#     fabric local regeneration beats any broadcast: moving 7/8 of a 140 GB
+#     model over xGMI costs ~40x more time than writing it locally. This is
+#     the MI355X-native default for DDP-style replication.
+#
+#   * "shard" — tensors are assigned to ranks (greedy size-balanced,
+#     deterministic); each rank materializes only the tape subgraphs of its
+#     own tensors (the per-tensor call-stack replay makes the untouched
+#     subgraphs free). Non-owned entries stay fake. This is the FSDP-style
+#     init path: N ranks do 1/N of the HBM writes each, no communication.
+#
+#   * "broadcast" — shard ownership as above, but every rank ends with the
+#     full module: owners materialize their tensors while non-owners
+#     allocate, and each tensor is broadcast from its owner with async
+#     collectives so the RCCL transfers overlap with the init kernels of
+#     the tensors still being materialized. Use when ranks cannot be
+#     trusted to share RNG state (e.g. mixed seeds) but must end
+#     bitwise-identical.
+
+from typing import Callable, Dict, List, Optional, Tuple
+
+import torch
+import torch.distributed as dist
+from torch.nn import Module, Parameter
+
+from torchdistx_amd import _C
+from torchdistx_amd.deferred_init import _restore_class
+
+
+def _named_deferred_tensors(
+    module: Module, buffers_only: bool
+) -> List[Tuple[Module, str, torch.Tensor, bool]]:
+    """(owner module, attribute key, tensor, is_param) for every deferred
+    parameter/buffer, in deterministic module-traversal order."""
+    out = []
+    for submodule in module.modules():
+        if not buffers_only:
+            for key, p in submodule._parameters.items():
+                if p is not None and _C.can_materialize(p):
+                    out.append((submodule, key, p, True))
+        for key, b in submodule._buffers.items():
+            if b is not None and _C.can_materialize(b):
+                out.append((submodule, key, b, False))
+    return out
+
+
+def assign_owners(sizes: List[int], world_size: int) -> List[int]:
+    """Greedy size-balanced deterministic assignment: largest tensors
+    first, each to the currently least-loaded rank."""
+    loads = [0] * world_size
+    owners = [0] * len(sizes)
+    order = sorted(range(len(sizes)), key=lambda i: (-sizes[i], i))
+    for i in order:
+        rank = min(range(world_size), key=lambda r: (loads[r], r))
+        owners[i] = rank
+        loads[rank] += sizes[i]
+    return owners
+
+
+def materialize_module_distributed(
+    module: Module,
+    mode: str = "replicate",
+    process_group: Optional[dist.ProcessGroup] = None,
+    buffers_only: bool = False,
+    check_fn: Optional[Callable[[Module], bool]] = None,
+) -> Dict[str, int]:
+    """Materializes a deferred ``module`` across the ranks of
+    ``process_group`` (default: the world). Returns {tensor index -> owner
+    rank} for the shard/broadcast modes ({} for replicate).
+
+    See the module docstring for the mode semantics.
+    """
+    if mode not in ("replicate", "shard", "broadcast"):
+        raise ValueError(f"unknown materialization mode: {mode!r}")
+
+    if mode == "replicate" or not dist.is_initialized():
+        from torchdistx_amd.deferred_init import materialize_module
+
+        materialize_module(module, buffers_only=buffers_only, check_fn=check_fn)
+        return {}
+
+    group = process_group or dist.group.WORLD
+    world = dist.get_world_size(group)
+    rank = dist.get_rank(group)
+
+    entries = _named_deferred_tensors(module, buffers_only)
+    if check_fn is not None:
+        entries = [e for e in entries if check_fn(e[0])]
+    owners = assign_owners([t.numel() for _, _, t, _ in entries], world)
+
+    handles = []
+    for i, (submodule, key, tensor, is_param) in enumerate(entries):
+        owner = owners[i]
+        if rank == owner:
+            mat = _C.materialize_tensor(tensor)
+        elif mode == "shard":
+            continue
+        else:
+            mat = torch.empty(
+                tensor.shape, dtype=tensor.dtype, device=tensor.device
+            )
+            if is_param:
+                mat.requires_grad_(tensor.requires_grad)
+        if mode == "broadcast":
+            src = dist.get_global_rank(group, owner)
+            # async_op lets RCCL run this transfer on its communication
+            # stream while the default stream keeps launching init kernels
+            # for the tensors this rank still owns. detach(): collectives
+            # reject autograd-tracked tensors; the storage is shared.
+            handles.append(
+                dist.broadcast(mat.detach(), src=src, group=group, async_op=True)
+            )
+        if is_param:
+            mat = _restore_class(tensor, mat)
+            submodule._parameters[key] = mat
+        else:
+            submodule._buffers[key] = mat
+
+    for h in handles:
+        h.wait()
+    return {i: owners[i] for i in range(len(entries))}
