@@ -24,6 +24,7 @@
 #include <unordered_set>
 #include <vector>
 
+#include <ATen/CPUGeneratorImpl.h>
 #include <ATen/Context.h>
 #include <ATen/ThreadLocalState.h>
 #include <ATen/core/dispatch/Dispatcher.h>
@@ -76,21 +77,53 @@ struct RecordedOp {
 // the native CDNA4 kernels run the replay).
 constexpr uint64_t kPhiloxStridePerOp = 4;
 
-std::atomic<uint64_t> next_rng_slot{1};
+// Session state: the seed is a 64-bit nonce drawn from the default CPU
+// generator when the outermost deferred_init is entered, and slots count
+// from 1 within the session. Consequences (all intentional):
+//   * torch.manual_seed(S) before deferred_init fixes the init bits;
+//   * two deferred_init sessions without re-seeding produce different
+//     inits (the second nonce differs), matching eager expectations;
+//   * within one tape, replaying any subset on any rank gives identical
+//     bits (partition-invariant sharded materialization).
+std::atomic<uint64_t> session_rng_slot{1};
+std::atomic<uint64_t> session_rng_seed{0};
+std::atomic<bool> session_nonce_drawn{false};
 
 bool isRngOpName(const std::string& name) {
   return name == "aten::uniform_" || name == "aten::normal_";
 }
 
+void beginRngSession() {
+  // Lazy: the nonce is drawn from the CPU generator only when the first
+  // GPU-targeted RNG op is recorded, so pure-CPU tapes never perturb the
+  // generator stream and their replay stays bitwise-equal to eager init.
+  session_nonce_drawn.store(false, std::memory_order_relaxed);
+  session_rng_slot.store(1, std::memory_order_relaxed);
+}
+
+uint64_t sessionNonce() {
+  if (!session_nonce_drawn.exchange(true, std::memory_order_relaxed)) {
+    auto gen = at::globalContext().defaultGenerator(c10::DeviceType::CPU);
+    uint64_t nonce;
+    {
+      std::lock_guard<std::mutex> lock(gen.mutex());
+      nonce = at::check_generator<at::CPUGeneratorImpl>(gen)->random64();
+    }
+    session_rng_seed.store(nonce, std::memory_order_relaxed);
+  }
+  return session_rng_seed.load(std::memory_order_relaxed);
+}
+
+// Pins (seed, counter-offset) for an RNG op whose target lives on the GPU
+// (where the tdx Philox kernels consume it). CPU-targeted RNG ops replay
+// through the stock generator to preserve eager bit-parity.
 std::optional<std::pair<uint64_t, uint64_t>> pinPhiloxForOp(
-    const std::string& name) {
-  if (!isRngOpName(name)) {
+    const std::string& name, bool cuda_target) {
+  if (!cuda_target || !isRngOpName(name)) {
     return std::nullopt;
   }
-  uint64_t seed = at::globalContext()
-                      .defaultGenerator(c10::DeviceType::CPU)
-                      .current_seed();
-  uint64_t slot = next_rng_slot.fetch_add(1, std::memory_order_relaxed);
+  uint64_t seed = sessionNonce();
+  uint64_t slot = session_rng_slot.fetch_add(1, std::memory_order_relaxed);
   return std::make_pair(seed, slot * kPhiloxStridePerOp);
 }
 
@@ -303,7 +336,13 @@ void deferredInitHandler(const c10::OperatorHandle& op,
   });
 
   if (has_fake_arg || has_fake_ret) {
-    auto philox = pinPhiloxForOp(schema.operator_name().name);
+    bool cuda_target = false;
+    visitTensors(*stack, rets_begin, stack->size(), [&](const at::Tensor& t) {
+      if (auto* fake = asFake(t)) {
+        cuda_target = cuda_target || fake->fake_device().is_cuda();
+      }
+    });
+    auto philox = pinPhiloxForOp(schema.operator_name().name, cuda_target);
     recordOp(
         schema.operator_name().name,
         [handle = op, philox](torch::jit::Stack& s) {
@@ -693,6 +732,7 @@ void replayNode(const std::shared_ptr<OpNode>& node) {
 
 void enterDeferredInit() {
   if (deferred_level++ == 0) {
+    beginRngSession();
     c10::impl::tls_set_dispatch_key_included(kDeferredKey, true);
     installProxyHooks();
   }
