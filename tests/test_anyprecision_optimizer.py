@@ -43,6 +43,9 @@ def test_fp32_no_kahan_matches_adamw_bitwise(device) -> None:
         use_kahan_summation=False,
         momentum_dtype=torch.float32,
         variance_dtype=torch.float32,
+        # Bitwise parity is an eager-path property; the fused CDNA4 kernel
+        # is checked against eager with tolerances in test_gpu.py.
+        use_fused=False,
     )
 
     for step in range(6):

@@ -46,6 +46,11 @@ class AnyPrecisionAdamW(Optimizer):
         variance_dtype: dtype of ``exp_avg_sq`` (default: torch.bfloat16).
         compensation_buffer_dtype: dtype of the Kahan compensation buffer
             (default: torch.bfloat16).
+        use_fused: force (True) or forbid (False) the fused single-kernel
+            CDNA4 step; None (default) uses it automatically on GPU
+            tensors when the kernel extension is loaded. The fused step
+            computes in fp32 and rounds once per state store, so it is
+            close to but not bitwise-identical with the eager op sequence.
     """
 
     def __init__(
@@ -59,6 +64,7 @@ class AnyPrecisionAdamW(Optimizer):
         momentum_dtype=torch.float32,
         variance_dtype=torch.bfloat16,
         compensation_buffer_dtype=torch.bfloat16,
+        use_fused=None,
     ):
         defaults = dict(
             lr=lr,
@@ -71,6 +77,7 @@ class AnyPrecisionAdamW(Optimizer):
             compensation_buffer_dtype=compensation_buffer_dtype,
         )
         super().__init__(params, defaults)
+        self.use_fused = use_fused
 
     @torch.no_grad()
     def step(self, closure=None):
@@ -117,7 +124,12 @@ class AnyPrecisionAdamW(Optimizer):
                 bias_correction2_sqrt = (1 - beta2**step) ** 0.5
                 step_size = lr / bias_correction1
 
-                if _fused_step_available(p):
+                fused_ok = (
+                    self.use_fused
+                    if self.use_fused is not None
+                    else _fused_step_available(p)
+                )
+                if fused_ok and _fused_step_available(p):
                     from torchdistx_amd import _kernels
 
                     _kernels.anyprecision_adamw_(
