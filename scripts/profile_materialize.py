@@ -1,4 +1,5 @@
 """Phase breakdown + cProfile of one deferred_init->materialize step on GPU."""
+import sys, os; sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import cProfile
 import pstats
 import sys

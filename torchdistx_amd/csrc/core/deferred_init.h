@@ -42,4 +42,19 @@ bool canMaterialize(const at::Tensor& tensor) noexcept;
 // fake tensor with no record.
 at::Tensor materializeTensor(const at::Tensor& tensor);
 
+// Introspection of a deferred tensor's tape record (observability beyond
+// the reference, which exposes nothing). All fields are cheap to compute
+// except pending_ops, which runs the same call-stack analysis
+// materialization would.
+struct RecordInfo {
+  uint64_t op_nr = 0;          // tape position of the producing op
+  size_t output_index = 0;     // which output of that op
+  bool materialized = false;   // producer already replayed?
+  std::string op_name;         // producer op (empty once replayed)
+  size_t pending_ops = 0;      // ops a materialize call would replay now
+};
+
+// std::nullopt when `tensor` carries no deferred-init record.
+std::optional<RecordInfo> recordInfo(const at::Tensor& tensor);
+
 }  // namespace tdx
