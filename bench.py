@@ -21,7 +21,6 @@ Launch (driver contract):
 """
 
 import argparse
-import gc
 import json
 import os
 import resource
@@ -93,8 +92,9 @@ def main():
         materialize_module_distributed(module, mode=args.mode)
         if args.mode != "shard":
             assert not is_deferred(module), "materialization incomplete"
+        # Reference-counting frees the module and its tape immediately (the
+        # tape's ownership graph is acyclic by design); no gc.collect().
         del module
-        gc.collect()
 
     def barrier_sync():
         if distributed:

@@ -51,12 +51,24 @@ struct OpOutputDescriptor {
   size_t index = 0;  // tensor-visit-order index into the node's outputs
 };
 
+// All tensors sharing one (meta) storage share one AliasGroup, which owns
+// every node that wrote to that storage. Any surviving member of the alias
+// family therefore keeps the whole family's tape segments alive — a view
+// can outlive its base (and vice versa) without losing in-place writes —
+// and, unlike record<->record keep-alive edges, the ownership graph is
+// acyclic: groups own nodes, nodes own dependency nodes, records own their
+// group. When the last fake of the family dies, the group and its nodes
+// (including any materialized outputs they cache) are released.
+struct AliasGroup {
+  std::vector<std::shared_ptr<OpNode>> writers;
+};
+
 // Per-fake-tensor record, stored in the impl's dispatch-key side table.
-// Mutable: an in-place op restamps `desc` in place, so records held alive
-// through `keep_alive` (view relationships) keep following the tape.
+// Mutable: an in-place op restamps `desc` in place, so aliased tensors
+// observe the rewrite through their shared group.
 struct TensorRecord {
   OpOutputDescriptor desc;
-  std::vector<std::shared_ptr<TensorRecord>> keep_alive;
+  std::shared_ptr<AliasGroup> group;
 };
 
 // One recorded call frame.
@@ -236,20 +248,29 @@ void recordOp(std::string name,
       rec = std::make_shared<TensorRecord>();
       setRecord(fake, rec);
     }
-    // Restamp IN PLACE: records reached through view keep-alives observe
-    // the rewrite (reference's view_records_, deferred_init.cc:126-154).
+    // Restamp IN PLACE: aliased records observe the rewrite through the
+    // shared group (fills the role of the reference's view_records_,
+    // deferred_init.cc:126-154).
     rec->desc = OpOutputDescriptor{node, out_idx};
 
-    // View keep-alive wiring: when an output aliases a *different* fake
-    // input, tie the two records' lifetimes together so whichever side
-    // outlives the other still holds the tape segments touching the shared
-    // storage.
-    for (size_t i = 0; i < input_records.size(); ++i) {
-      if (input_records[i].get() != rec.get() &&
-          storagesAlias(out_storage, input_storages[i])) {
-        rec->keep_alive.push_back(input_records[i]);
-        input_records[i]->keep_alive.push_back(rec);
+    // Alias-group wiring: adopt the group of any aliased fake input (a
+    // view adopts its base's family), else keep/create our own, and
+    // register this node as a writer of the family's storage.
+    if (rec->group == nullptr) {
+      for (size_t i = 0; i < input_records.size(); ++i) {
+        if (storagesAlias(out_storage, input_storages[i]) &&
+            input_records[i]->group != nullptr) {
+          rec->group = input_records[i]->group;
+          break;
+        }
       }
+      if (rec->group == nullptr) {
+        rec->group = std::make_shared<AliasGroup>();
+      }
+    }
+    if (rec->group->writers.empty() ||
+        rec->group->writers.back() != node) {
+      rec->group->writers.push_back(node);
     }
   });
 }
@@ -762,6 +783,34 @@ NoDeferredInit::~NoDeferredInit() {
 bool canMaterialize(const at::Tensor& tensor) noexcept {
   auto* fake = asFake(tensor);
   return fake != nullptr && getRecord(fake) != nullptr;
+}
+
+std::optional<RecordInfo> recordInfo(const at::Tensor& tensor) {
+  auto* fake = asFake(tensor);
+  if (fake == nullptr) {
+    return std::nullopt;
+  }
+  auto rec = getRecord(fake);
+  if (rec == nullptr) {
+    return std::nullopt;
+  }
+  RecordInfo info;
+  const auto& node = rec->desc.node;
+  info.op_nr = node->op_nr;
+  info.output_index = rec->desc.index;
+  info.materialized = node->materialized;
+  if (node->op.has_value()) {
+    info.op_name = node->op->name;
+  }
+  if (!node->materialized) {
+    const c10::Storage& storage = node->output_storages.at(rec->desc.index);
+    for (const auto& n : buildCallStack(node, storage)) {
+      if (!n->materialized) {
+        ++info.pending_ops;
+      }
+    }
+  }
+  return info;
 }
 
 at::Tensor materializeTensor(const at::Tensor& tensor) {

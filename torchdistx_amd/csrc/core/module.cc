@@ -78,6 +78,20 @@ PYBIND11_MODULE(_C, m) {
         pybind11::arg("enabled"));
   m.def("native_init_enabled", &tdx::nativeInitEnabled);
 
+  m.def("record_info", [](const at::Tensor& t) -> pybind11::object {
+    auto info = tdx::recordInfo(t);
+    if (!info.has_value()) {
+      return pybind11::none();
+    }
+    pybind11::dict d;
+    d["op_nr"] = info->op_nr;
+    d["output_index"] = info->output_index;
+    d["materialized"] = info->materialized;
+    d["op_name"] = info->op_name;
+    d["pending_ops"] = info->pending_ops;
+    return d;
+  });
+
   m.def("materialize_tensor", [](const at::Tensor& t) {
     at::Tensor out;
     {
