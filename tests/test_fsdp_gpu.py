@@ -1,0 +1,76 @@
+# FSDP integration on GPU: the deferred_init -> FSDP(param_init_fn=...)
+# flow the feature exists for, plus the comm-hook registration path.
+# Multi-rank hook numerics are covered CPU-side in test_comm_hooks_multiproc
+# and test_gossip_grad (gloo); tests here follow the reference's
+# skip_if_lt_x_gpu pattern for >1-GPU cases.
+
+import os
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture()
+def nccl_world1():
+    if not torch.cuda.is_available():
+        pytest.skip("needs a ROCm GPU")
+    import torch.distributed as dist
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29551")
+    torch.cuda.set_device(0)
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    yield dist
+    dist.destroy_process_group()
+
+
+def test_fsdp_with_deferred_init_param_init_fn(nccl_world1) -> None:
+    from torch.distributed.fsdp import FullyShardedDataParallel as FSDP
+
+    from torchdistx_amd import deferred_init, materialize_module
+    from torchdistx_amd.models import TINY, build_model
+
+    torch.manual_seed(0)
+    module = deferred_init(
+        build_model, TINY, device="cuda", dtype=torch.float32
+    )
+    fsdp = FSDP(
+        module,
+        param_init_fn=lambda m: materialize_module(m),
+        device_id=torch.cuda.current_device(),
+    )
+    tokens = torch.randint(0, TINY.vocab_size, (2, 16), device="cuda")
+    loss = fsdp(tokens[:, :-1]).float().mean()
+    loss.backward()
+    torch.cuda.synchronize()
+    assert loss.isfinite().item()
+
+
+def test_slowmo_hook_registers_on_fsdp(nccl_world1) -> None:
+    from torch.distributed.fsdp import FullyShardedDataParallel as FSDP
+    from torch.distributed.fsdp import ShardingStrategy
+
+    from torchdistx_amd.slowmo import (
+        SlowMomentumOptimizer,
+        SlowMoState,
+        slowmo_hook,
+    )
+
+    dist = nccl_world1
+    net = torch.nn.Linear(8, 8).cuda()
+    fsdp = FSDP(net, sharding_strategy=ShardingStrategy.NO_SHARD)
+    state = SlowMoState(dist.group.WORLD, sync_grads=True)
+    fsdp.register_comm_hook(state, slowmo_hook)
+
+    optim = SlowMomentumOptimizer(
+        torch.optim.SGD(fsdp.parameters(), lr=0.1), slowmo_freq=2
+    )
+    for _ in range(3):
+        optim.zero_grad()
+        fsdp(torch.randn(4, 8, device="cuda")).square().mean().backward()
+        optim.step()
+    torch.cuda.synchronize()
+    sd = optim.state_dict()
+    assert sd["step"] == 3
