@@ -120,3 +120,40 @@ def test_state_dtypes_follow_config() -> None:
     assert state["exp_avg"].dtype == torch.bfloat16
     assert state["exp_avg_sq"].dtype == torch.bfloat16
     assert state["compensation"].dtype == torch.float16
+
+
+def test_state_dict_roundtrip() -> None:
+    torch.manual_seed(5)
+    model = torch.nn.Linear(6, 6)
+    opt = AnyPrecisionAdamW(model.parameters(), lr=1e-2, use_kahan_summation=True,
+                            momentum_dtype=torch.float32,
+                            variance_dtype=torch.float32)
+    for _ in range(3):
+        opt.zero_grad()
+        model(torch.randn(4, 6)).square().mean().backward()
+        opt.step()
+    # Serialize through a real checkpoint boundary (torch.save/load), which
+    # breaks the tensor aliasing a live state_dict keeps.
+    import io
+
+    buf = io.BytesIO()
+    torch.save(opt.state_dict(), buf)
+    buf.seek(0)
+    sd = torch.load(buf)
+
+    model2 = torch.nn.Linear(6, 6)
+    model2.load_state_dict(model.state_dict())
+    opt2 = AnyPrecisionAdamW(model2.parameters(), lr=1e-2, use_kahan_summation=True,
+                             momentum_dtype=torch.float32,
+                             variance_dtype=torch.float32)
+    opt2.load_state_dict(sd)
+
+    # One identical step after restore must produce identical params.
+    torch.manual_seed(77)
+    x = torch.randn(4, 6)
+    for m, o in ((model, opt), (model2, opt2)):
+        o.zero_grad()
+        m(x).square().mean().backward()
+        o.step()
+    for p1, p2 in zip(model.parameters(), model2.parameters()):
+        assert torch.equal(p1, p2)

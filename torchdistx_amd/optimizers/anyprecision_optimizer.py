@@ -79,6 +79,27 @@ class AnyPrecisionAdamW(Optimizer):
         super().__init__(params, defaults)
         self.use_fused = use_fused
 
+    def load_state_dict(self, state_dict):
+        """Base-class restore, then re-cast the optimizer states to their
+        configured dtypes: ``torch.optim.Optimizer.load_state_dict`` casts
+        floating state tensors to the *parameter* dtype, which would
+        silently destroy the any-precision state layout (e.g. a bf16
+        compensation buffer next to fp32 weights)."""
+        super().load_state_dict(state_dict)
+        for group in self.param_groups:
+            for p in group["params"]:
+                state = self.state.get(p)
+                if not state:
+                    continue
+                state["exp_avg"] = state["exp_avg"].to(group["momentum_dtype"])
+                state["exp_avg_sq"] = state["exp_avg_sq"].to(
+                    group["variance_dtype"]
+                )
+                if "compensation" in state:
+                    state["compensation"] = state["compensation"].to(
+                        group["compensation_buffer_dtype"]
+                    )
+
     @torch.no_grad()
     def step(self, closure=None):
         """Performs a single optimization step."""
