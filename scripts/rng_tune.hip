@@ -1,0 +1,162 @@
+// Standalone tuning probe for the Philox uniform kernel: sweeps
+// elements-per-thread and grid size on a 4 GiB bf16 buffer.
+// Build/run on the GPU box:
+//   hipcc --offload-arch=gfx950 -O3 -std=c++17 scripts/rng_tune.hip -o /tmp/rng_tune && /tmp/rng_tune
+#include <hip/hip_bf16.h>
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+#include <cstdio>
+
+#define CHECK(x)                                            \
+  do {                                                      \
+    hipError_t e = (x);                                     \
+    if (e != hipSuccess) {                                  \
+      printf("HIP error %s at %d\n", hipGetErrorString(e), __LINE__); \
+      return 1;                                             \
+    }                                                       \
+  } while (0)
+
+__device__ __forceinline__ uint2 mulhilo32(uint32_t a, uint32_t b) {
+  uint2 r;
+  r.x = a * b;
+  r.y = __umulhi(a, b);
+  return r;
+}
+
+__device__ __forceinline__ uint4 philox10(uint64_t seed, uint64_t subseq,
+                                          uint64_t offset) {
+  constexpr uint32_t kW0 = 0x9E3779B9u, kW1 = 0xBB67AE85u;
+  constexpr uint32_t kM0 = 0xD2511F53u, kM1 = 0xCD9E8D57u;
+  uint32_t k0 = (uint32_t)seed, k1 = (uint32_t)(seed >> 32);
+  uint4 c = make_uint4((uint32_t)offset, (uint32_t)(offset >> 32),
+                       (uint32_t)subseq, (uint32_t)(subseq >> 32));
+#pragma unroll
+  for (int r = 0; r < 10; ++r) {
+    uint2 r0 = mulhilo32(kM0, c.x);
+    uint2 r1 = mulhilo32(kM1, c.z);
+    c = make_uint4(r1.y ^ c.y ^ k0, r1.x, r0.y ^ c.w ^ k1, r0.x);
+    k0 += kW0;
+    k1 += kW1;
+  }
+  return c;
+}
+
+__device__ __forceinline__ float u2f(uint32_t x) {
+  return (float)(x >> 8) * (1.0f / 16777216.0f);
+}
+
+struct alignas(16) V8 {
+  __hip_bfloat16 v[8];
+};
+
+// EPT = elements per thread per grid-stride iteration (multiple of 8).
+template <int EPT>
+__global__ void rng_bf16(__hip_bfloat16* __restrict__ out, uint32_t n8,
+                         float a, float b, uint64_t seed, uint64_t offset) {
+  // n8 = number of 8-element groups; 32-bit indexing (n < 2^31 groups).
+  constexpr int G = EPT / 8;  // 16B stores per iteration
+  const uint32_t stride = gridDim.x * blockDim.x;
+  for (uint32_t g0 = blockIdx.x * blockDim.x + threadIdx.x; g0 * G < n8;
+       g0 += stride) {
+#pragma unroll
+    for (int s = 0; s < G; ++s) {
+      uint32_t g = g0 * G + s;
+      if (g >= n8) break;
+      float vals[8];
+#pragma unroll
+      for (int c = 0; c < 2; ++c) {
+        uint4 bits = philox10(seed, (uint64_t)g * 2 + c, offset);
+        vals[c * 4 + 0] = fmaf(u2f(bits.x), b, a);
+        vals[c * 4 + 1] = fmaf(u2f(bits.y), b, a);
+        vals[c * 4 + 2] = fmaf(u2f(bits.z), b, a);
+        vals[c * 4 + 3] = fmaf(u2f(bits.w), b, a);
+      }
+      V8 v;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v.v[j] = __float2bfloat16(vals[j]);
+      *reinterpret_cast<V8*>(out + (uint64_t)g * 8) = v;
+    }
+  }
+}
+
+// Contiguous-per-thread variant: thread handles G consecutive groups.
+template <int G>
+__global__ void rng_bf16_wide(__hip_bfloat16* __restrict__ out, uint32_t n8,
+                              float a, float b, uint64_t seed,
+                              uint64_t offset) {
+  const uint32_t stride = gridDim.x * blockDim.x;
+  for (uint32_t t = blockIdx.x * blockDim.x + threadIdx.x; t * G < n8;
+       t += stride) {
+    uint4 bits[2 * G];
+#pragma unroll
+    for (int c = 0; c < 2 * G; ++c) {
+      uint32_t g = t * G + c / 2;
+      bits[c] = philox10(seed, (uint64_t)g * 2 + (c & 1), offset);
+    }
+#pragma unroll
+    for (int s = 0; s < G; ++s) {
+      uint32_t g = t * G + s;
+      if (g >= n8) break;
+      V8 v;
+#pragma unroll
+      for (int c = 0; c < 2; ++c) {
+        uint4 bb = bits[s * 2 + c];
+        v.v[c * 4 + 0] = __float2bfloat16(fmaf(u2f(bb.x), b, a));
+        v.v[c * 4 + 1] = __float2bfloat16(fmaf(u2f(bb.y), b, a));
+        v.v[c * 4 + 2] = __float2bfloat16(fmaf(u2f(bb.z), b, a));
+        v.v[c * 4 + 3] = __float2bfloat16(fmaf(u2f(bb.w), b, a));
+      }
+      *reinterpret_cast<V8*>(out + (uint64_t)g * 8) = v;
+    }
+  }
+}
+
+template <typename K>
+double bench(K kernel, __hip_bfloat16* buf, uint32_t n8, int blocks,
+             int iters) {
+  hipEvent_t e0, e1;
+  hipEventCreate(&e0);
+  hipEventCreate(&e1);
+  kernel<<<blocks, 256>>>(buf, n8, -1.f, 2.f, 42, 4);  // warm
+  hipDeviceSynchronize();
+  hipEventRecord(e0);
+  for (int i = 0; i < iters; ++i) {
+    kernel<<<blocks, 256>>>(buf, n8, -1.f, 2.f, 42, 4);
+  }
+  hipEventRecord(e1);
+  hipEventSynchronize(e1);
+  float ms;
+  hipEventElapsedTime(&ms, e0, e1);
+  double bytes = (double)n8 * 16.0 * iters;
+  return bytes / (ms / 1e3) / 1e12;  // TB/s
+}
+
+int main() {
+  const uint64_t bytes = 4ull << 30;
+  const uint32_t n8 = (uint32_t)(bytes / 16);
+  __hip_bfloat16* buf;
+  CHECK(hipMalloc(&buf, bytes));
+
+  for (int blocks : {2048, 4096, 8192}) {
+    printf("blocks=%d\n", blocks);
+    printf("  strided EPT8 : %.2f TB/s\n", bench(rng_bf16<8>, buf, n8, blocks, 5));
+    printf("  strided EPT16: %.2f TB/s\n", bench(rng_bf16<16>, buf, n8, blocks, 5));
+    printf("  wide G2      : %.2f TB/s\n", bench(rng_bf16_wide<2>, buf, n8, blocks, 5));
+    printf("  wide G4      : %.2f TB/s\n", bench(rng_bf16_wide<4>, buf, n8, blocks, 5));
+    printf("  wide G8      : %.2f TB/s\n", bench(rng_bf16_wide<8>, buf, n8, blocks, 5));
+  }
+  // memset reference ceiling
+  hipEvent_t e0, e1;
+  hipEventCreate(&e0);
+  hipEventCreate(&e1);
+  hipEventRecord(e0);
+  for (int i = 0; i < 5; ++i) hipMemsetAsync(buf, 0, bytes);
+  hipEventRecord(e1);
+  hipEventSynchronize(e1);
+  float ms;
+  hipEventElapsedTime(&ms, e0, e1);
+  printf("hipMemset      : %.2f TB/s\n", (double)bytes * 5 / (ms / 1e3) / 1e12);
+  hipFree(buf);
+  return 0;
+}

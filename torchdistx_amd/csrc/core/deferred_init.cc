@@ -142,6 +142,13 @@ std::optional<std::pair<uint64_t, uint64_t>> pinPhiloxForOp(
 struct InputSlot {
   std::optional<OpOutputDescriptor> desc;  // set when the arg was fake
   std::optional<uint32_t> external_version;  // set for real tensor args
+  // For fake inputs read from a *different* storage family than this op
+  // writes: a record-time snapshot of that family's writers so far. Replay
+  // of this op must first replay these (they produced the value the op
+  // observed), and holding them here keeps cross-family history alive even
+  // after the other family's fakes are gone. Acyclic by construction:
+  // every held node has a smaller op_nr.
+  std::vector<std::shared_ptr<OpNode>> history;
 };
 
 struct OpNode {
@@ -201,6 +208,7 @@ void recordOp(std::string name,
   // for external (real) tensor args.
   std::vector<std::shared_ptr<TensorRecord>> input_records;
   std::vector<c10::Storage> input_storages;
+  std::vector<size_t> fake_slot_idx;  // input_slots index per fake input
   {
     torch::jit::Stack args_stack{args.begin(), args.end()};
     mapTensors(args_stack, 0, args_stack.size(),
@@ -212,7 +220,8 @@ void recordOp(std::string name,
                        "A fake tensor constructed outside of a deferred-init "
                        "context (or stripped of its record) was used in `",
                        name, "` and cannot be recorded.");
-                   node->input_slots.push_back(InputSlot{rec->desc, {}});
+                   fake_slot_idx.push_back(node->input_slots.size());
+                   node->input_slots.push_back(InputSlot{rec->desc, {}, {}});
                    rec->desc.node->dependents.emplace_back(node);
                    input_records.push_back(std::move(rec));
                    input_storages.push_back(fake->meta_tensor().storage());
@@ -220,7 +229,7 @@ void recordOp(std::string name,
                  }
                  if (t.defined()) {
                    node->input_slots.push_back(
-                       InputSlot{{}, tensorVersion(t)});
+                       InputSlot{{}, tensorVersion(t), {}});
                  } else {
                    node->input_slots.push_back(InputSlot{});
                  }
@@ -273,6 +282,23 @@ void recordOp(std::string name,
       rec->group->writers.push_back(node);
     }
   });
+
+  // History snapshots: for every fake input whose storage this op does NOT
+  // write (a pure read from another family), capture that family's writer
+  // list as of now.
+  for (size_t i = 0; i < input_records.size(); ++i) {
+    bool writes_it = false;
+    for (const c10::Storage& out : node->output_storages) {
+      if (storagesAlias(out, input_storages[i])) {
+        writes_it = true;
+        break;
+      }
+    }
+    if (!writes_it && input_records[i]->group != nullptr) {
+      node->input_slots[fake_slot_idx[i]].history =
+          input_records[i]->group->writers;
+    }
+  }
 }
 
 bool isTerminalOp(const c10::OperatorHandle& op) {
@@ -580,31 +606,17 @@ bool sharesOutputStorage(const OpNode& a, const OpNode& b) {
   return false;
 }
 
-// Chronologically last node (transitively reachable through dependents)
-// whose outputs alias `storage` (reference getLastInPlaceOpNode,
-// deferred_init.cc:541-579).
-std::shared_ptr<OpNode> findLastAliasWriter(
-    const std::shared_ptr<OpNode>& start, const c10::Storage& storage) {
-  std::shared_ptr<OpNode> last = start;
-  std::unordered_set<OpNode*> visited{start.get()};
-  std::vector<std::shared_ptr<OpNode>> frontier{start};
-  while (!frontier.empty()) {
-    auto n = std::move(frontier.back());
-    frontier.pop_back();
-    for (const auto& wd : n->dependents) {
-      auto d = wd.lock();
-      if (!d || !visited.insert(d.get()).second) {
-        continue;
-      }
-      if (d->op_nr > last->op_nr && outputsAlias(*d, storage)) {
-        last = d;
-      }
-      frontier.push_back(std::move(d));
-    }
-  }
-  return last;
-}
-
+// The ordered replay set for the tensor behind `rec` (fills the role of
+// the reference's buildCallStack + getLastInPlaceOpNode,
+// deferred_init.cc:506-622, but is driven by the alias group instead of a
+// dependents walk — a dependents walk cannot reach in-place writes made to
+// a *base* after a view was taken, since those hang off the base's
+// producer, not the view node):
+//   1. every writer of the target's storage family (from the AliasGroup),
+//      with their dependency closures and cross-family history snapshots;
+//   2. to fixpoint, dependents of collected nodes whose inputs a collected
+//      later writer would clobber — they must replay now, in chronological
+//      position, or they would later observe post-overwrite values.
 struct CallStack {
   std::vector<std::shared_ptr<OpNode>> nodes;
   std::unordered_set<OpNode*> members;
@@ -618,24 +630,25 @@ struct CallStack {
       if (slot.desc.has_value()) {
         addWithDeps(slot.desc->node);
       }
+      for (const auto& h : slot.history) {
+        addWithDeps(h);
+      }
     }
   }
 };
 
-// Builds the ordered replay set for `target` (reference buildCallStack +
-// collectCallStack, deferred_init.cc:530-622): the dependency closure of
-// the target and of the last writer aliasing the target's storage, plus —
-// to fixpoint — (a) dependents up to that writer that share storage with a
-// collected node (in-place writers and views of the same storage) and
-// (b) dependents whose inputs a later collected writer would clobber,
-// which therefore must replay now or never see the right values.
-std::vector<std::shared_ptr<OpNode>> buildCallStack(
-    const std::shared_ptr<OpNode>& target, const c10::Storage& storage) {
-  auto last = findLastAliasWriter(target, storage);
-
+std::vector<std::shared_ptr<OpNode>> buildCallStack(const TensorRecord& rec) {
   CallStack cs;
-  cs.addWithDeps(target);
-  cs.addWithDeps(last);
+  uint64_t horizon = 0;
+  if (rec.group != nullptr) {
+    for (const auto& w : rec.group->writers) {
+      horizon = std::max(horizon, w->op_nr);
+      cs.addWithDeps(w);
+    }
+  } else {
+    cs.addWithDeps(rec.desc.node);
+    horizon = rec.desc.node->op_nr;
+  }
 
   bool changed = true;
   while (changed) {
@@ -645,35 +658,32 @@ std::vector<std::shared_ptr<OpNode>> buildCallStack(
     for (const auto& n : snapshot) {
       for (const auto& wd : n->dependents) {
         auto d = wd.lock();
-        if (!d || cs.members.count(d.get()) != 0 || d->materialized) {
+        if (!d || cs.members.count(d.get()) != 0 || d->materialized ||
+            d->op_nr > horizon) {
           continue;
         }
+        // Clobber rule: `d` reads an output of a collected node whose
+        // storage a collected later op overwrites.
         bool include = false;
-        if (d->op_nr <= last->op_nr && sharesOutputStorage(*d, *n)) {
-          include = true;
-        } else {
-          // Clobber rule: `d` reads an output of a collected node whose
-          // storage a collected later op overwrites.
-          for (const InputSlot& slot : d->input_slots) {
-            if (!slot.desc.has_value() ||
-                cs.members.count(slot.desc->node.get()) == 0) {
-              continue;
-            }
-            const c10::Storage& s =
-                slot.desc->node->output_storages[slot.desc->index];
-            if (!s) {
-              continue;
-            }
-            for (const auto& w : cs.nodes) {
-              if (w->op_nr > d->op_nr && w.get() != slot.desc->node.get() &&
-                  outputsAlias(*w, s)) {
-                include = true;
-                break;
-              }
-            }
-            if (include) {
+        for (const InputSlot& slot : d->input_slots) {
+          if (!slot.desc.has_value() ||
+              cs.members.count(slot.desc->node.get()) == 0) {
+            continue;
+          }
+          const c10::Storage& s =
+              slot.desc->node->output_storages[slot.desc->index];
+          if (!s) {
+            continue;
+          }
+          for (const auto& w : cs.nodes) {
+            if (w->op_nr > d->op_nr && w.get() != slot.desc->node.get() &&
+                outputsAlias(*w, s)) {
+              include = true;
               break;
             }
+          }
+          if (include) {
+            break;
           }
         }
         if (include) {
@@ -803,8 +813,7 @@ std::optional<RecordInfo> recordInfo(const at::Tensor& tensor) {
     info.op_name = node->op->name;
   }
   if (!node->materialized) {
-    const c10::Storage& storage = node->output_storages.at(rec->desc.index);
-    for (const auto& n : buildCallStack(node, storage)) {
+    for (const auto& n : buildCallStack(*rec)) {
       if (!n->materialized) {
         ++info.pending_ops;
       }
@@ -824,8 +833,7 @@ at::Tensor materializeTensor(const at::Tensor& tensor) {
                     "so it cannot be materialized.");
 
   auto node = rec->desc.node;
-  const c10::Storage& storage = node->output_storages.at(rec->desc.index);
-  for (const auto& n : buildCallStack(node, storage)) {
+  for (const auto& n : buildCallStack(*rec)) {
     replayNode(n);
   }
 
