@@ -27,6 +27,8 @@
 #include <ATen/hip/HIPGeneratorImpl.h>
 #include <torch/library.h>
 
+#include <limits>
+
 #include <hip/hip_bf16.h>
 #include <hip/hip_fp16.h>
 
@@ -34,7 +36,11 @@ namespace tdx {
 namespace {
 
 constexpr int kBlock = 256;
-constexpr int kMaxBlocks = 2048;  // 256 CUs x 8 resident blocks
+// Measured on gfx950 (scripts/rng_tune.hip): this ALU-heavy streaming
+// kernel keeps improving past the usual ~2048-block guideline; 8192 blocks
+// + 32-bit indexing reaches 3.2 TB/s on a 4 GiB bf16 fill (pure-store
+// ceiling: 6.35 TB/s via hipMemset).
+constexpr int kMaxBlocks = 8192;
 
 // ---------------------------------------------------------------------------
 // Philox4x32-10 (standard constants), producing 4 x uint32 per invocation.
@@ -76,6 +82,13 @@ __device__ __forceinline__ uint4 philox10(uint64_t seed,
 // PyTorch's uniform transformation).
 __device__ __forceinline__ float u32_to_uniform(uint32_t x) {
   return static_cast<float>(x >> 8) * (1.0f / 16777216.0f);
+}
+
+// uint16 -> [0, 1) float. For 16-bit output dtypes (8-bit bf16 / 11-bit
+// fp16 mantissas) 16 random bits per sample are ample, and they halve the
+// Philox work per byte: one philox10 call yields a full 16-byte store.
+__device__ __forceinline__ float u16_to_uniform(uint32_t x) {
+  return static_cast<float>(x & 0xffffu) * (1.0f / 65536.0f);
 }
 
 // Two uniforms -> two standard normals (Box-Muller).
@@ -125,13 +138,15 @@ struct VecTraits<__half> {
 
 enum class Dist { kUniform, kNormal };
 
-// One kernel for both distributions: generates VecTraits<T>::kElems values
-// per iteration per thread with counter = the element-group index, applies
-// the affine transform (a + b*u for uniform, mean + std*n for normal), and
-// writes one 16-byte vector per iteration.
-template <typename T, Dist kDist>
+// One kernel for both distributions: each iteration runs ONE philox10 with
+// counter = the 16-byte group index and turns its 128 bits into one
+// 16-byte store — 4 fp32 samples (24-bit uniforms) or 8 bf16/fp16 samples
+// (16-bit uniforms, matched to the output mantissa). IdxT is uint32_t
+// whenever the group count fits (the hot case; 64-bit address math costs
+// ~2x VALU on this loop).
+template <typename T, Dist kDist, typename IdxT>
 __global__ void rng_kernel(T* __restrict__ out,
-                           int64_t n,
+                           IdxT n,
                            float a,
                            float b,
                            uint64_t seed,
@@ -139,34 +154,49 @@ __global__ void rng_kernel(T* __restrict__ out,
   constexpr int kElems = VecTraits<T>::kElems;
   using Vec = typename VecTraits<T>::Vec;
 
-  const int64_t n_groups = (n + kElems - 1) / kElems;
-  const int64_t stride = static_cast<int64_t>(gridDim.x) * blockDim.x;
+  const IdxT n_groups = (n + kElems - 1) / kElems;
+  const IdxT stride = static_cast<IdxT>(gridDim.x) * blockDim.x;
 
-  for (int64_t g = blockIdx.x * static_cast<int64_t>(blockDim.x) + threadIdx.x;
+  for (IdxT g = blockIdx.x * static_cast<IdxT>(blockDim.x) + threadIdx.x;
        g < n_groups; g += stride) {
+    uint4 bits = philox10(seed, static_cast<uint64_t>(g), offset);
     float vals[kElems];
-#pragma unroll
-    for (int c = 0; c < kElems / 4; ++c) {
-      uint4 bits = philox10(seed, static_cast<uint64_t>(g) * (kElems / 4) + c,
-                            offset);
+    if constexpr (kElems == 4) {
       float u[4] = {u32_to_uniform(bits.x), u32_to_uniform(bits.y),
                     u32_to_uniform(bits.z), u32_to_uniform(bits.w)};
       if constexpr (kDist == Dist::kUniform) {
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
-          vals[c * 4 + j] = fmaf(u[j], b, a);
+          vals[j] = fmaf(u[j], b, a);
         }
       } else {
         float2 n01 = box_muller(u[0], u[1]);
         float2 n23 = box_muller(u[2], u[3]);
-        vals[c * 4 + 0] = fmaf(n01.x, b, a);
-        vals[c * 4 + 1] = fmaf(n01.y, b, a);
-        vals[c * 4 + 2] = fmaf(n23.x, b, a);
-        vals[c * 4 + 3] = fmaf(n23.y, b, a);
+        vals[0] = fmaf(n01.x, b, a);
+        vals[1] = fmaf(n01.y, b, a);
+        vals[2] = fmaf(n23.x, b, a);
+        vals[3] = fmaf(n23.y, b, a);
+      }
+    } else {
+      uint32_t words[4] = {bits.x, bits.y, bits.z, bits.w};
+      if constexpr (kDist == Dist::kUniform) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          vals[j * 2 + 0] = fmaf(u16_to_uniform(words[j]), b, a);
+          vals[j * 2 + 1] = fmaf(u16_to_uniform(words[j] >> 16), b, a);
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          float2 nj = box_muller(u16_to_uniform(words[j]),
+                                 u16_to_uniform(words[j] >> 16));
+          vals[j * 2 + 0] = fmaf(nj.x, b, a);
+          vals[j * 2 + 1] = fmaf(nj.y, b, a);
+        }
       }
     }
 
-    const int64_t base = g * kElems;
+    const IdxT base = g * kElems;
     if (base + kElems <= n) {
       Vec v;
       T* vp = reinterpret_cast<T*>(&v);
@@ -176,7 +206,7 @@ __global__ void rng_kernel(T* __restrict__ out,
       }
       *reinterpret_cast<Vec*>(out + base) = v;
     } else {
-      for (int64_t j = 0; base + j < n; ++j) {
+      for (IdxT j = 0; base + j < n; ++j) {
         out[base + j] = from_float<T>(vals[j]);
       }
     }
@@ -265,10 +295,19 @@ void launchRng(at::Tensor& self,
     using T = decltype(type_tag);
     const int64_t n_groups =
         (n + VecTraits<T>::kElems - 1) / VecTraits<T>::kElems;
-    hipLaunchKernelGGL((rng_kernel<T, kDist>), dim3(numBlocks(n_groups)),
-                       dim3(kBlock), 0, stream.stream(),
-                       reinterpret_cast<T*>(self.data_ptr()), n, a, b, seed,
-                       offset);
+    if (n <= std::numeric_limits<uint32_t>::max() / 2) {
+      hipLaunchKernelGGL((rng_kernel<T, kDist, uint32_t>),
+                         dim3(numBlocks(n_groups)), dim3(kBlock), 0,
+                         stream.stream(),
+                         reinterpret_cast<T*>(self.data_ptr()),
+                         static_cast<uint32_t>(n), a, b, seed, offset);
+    } else {
+      hipLaunchKernelGGL((rng_kernel<T, kDist, uint64_t>),
+                         dim3(numBlocks(n_groups)), dim3(kBlock), 0,
+                         stream.stream(),
+                         reinterpret_cast<T*>(self.data_ptr()),
+                         static_cast<uint64_t>(n), a, b, seed, offset);
+    }
     C10_HIP_KERNEL_LAUNCH_CHECK();
   };
   switch (self.scalar_type()) {
