@@ -112,6 +112,78 @@ __global__ void rng_bf16_wide(__hip_bfloat16* __restrict__ out, uint32_t n8,
   }
 }
 
+__device__ __forceinline__ float2 bm(float u1, float u2) {
+  u1 = fmaxf(u1, 1.1754944e-38f);
+  float r = sqrtf(-2.0f * __logf(u1));
+  float sn, cs;
+  __sincosf(6.2831853071795865f * u2, &sn, &cs);
+  return make_float2(r * cs, r * sn);
+}
+
+// normal bf16: ONE philox -> 8 samples via 16-bit uniforms (kernel v2).
+__global__ void normal_bf16_u16(__hip_bfloat16* __restrict__ out, uint32_t n8,
+                                float a, float b, uint64_t seed,
+                                uint64_t offset) {
+  const uint32_t stride = gridDim.x * blockDim.x;
+  for (uint32_t g = blockIdx.x * blockDim.x + threadIdx.x; g < n8;
+       g += stride) {
+    uint4 bits = philox10(seed, g, offset);
+    uint32_t w[4] = {bits.x, bits.y, bits.z, bits.w};
+    V8 v;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float2 nj = bm((float)(w[j] & 0xffffu) * (1.0f / 65536.0f),
+                     (float)(w[j] >> 16) * (1.0f / 65536.0f));
+      v.v[j * 2 + 0] = __float2bfloat16(fmaf(nj.x, b, a));
+      v.v[j * 2 + 1] = __float2bfloat16(fmaf(nj.y, b, a));
+    }
+    *reinterpret_cast<V8*>(out + (uint64_t)g * 8) = v;
+  }
+}
+
+// normal bf16 with TWO philox and 24-bit uniforms (kernel v1 style).
+__global__ void normal_bf16_u32(__hip_bfloat16* __restrict__ out, uint32_t n8,
+                                float a, float b, uint64_t seed,
+                                uint64_t offset) {
+  const uint32_t stride = gridDim.x * blockDim.x;
+  for (uint32_t g = blockIdx.x * blockDim.x + threadIdx.x; g < n8;
+       g += stride) {
+    V8 v;
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      uint4 bits = philox10(seed, (uint64_t)g * 2 + c, offset);
+      float2 n01 = bm(u2f(bits.x), u2f(bits.y));
+      float2 n23 = bm(u2f(bits.z), u2f(bits.w));
+      v.v[c * 4 + 0] = __float2bfloat16(fmaf(n01.x, b, a));
+      v.v[c * 4 + 1] = __float2bfloat16(fmaf(n01.y, b, a));
+      v.v[c * 4 + 2] = __float2bfloat16(fmaf(n23.x, b, a));
+      v.v[c * 4 + 3] = __float2bfloat16(fmaf(n23.y, b, a));
+    }
+    *reinterpret_cast<V8*>(out + (uint64_t)g * 8) = v;
+  }
+}
+
+// uniform bf16 v2: ONE philox -> 8 samples via 16-bit uniforms.
+__global__ void uniform_bf16_u16(__hip_bfloat16* __restrict__ out,
+                                 uint32_t n8, float a, float b, uint64_t seed,
+                                 uint64_t offset) {
+  const uint32_t stride = gridDim.x * blockDim.x;
+  for (uint32_t g = blockIdx.x * blockDim.x + threadIdx.x; g < n8;
+       g += stride) {
+    uint4 bits = philox10(seed, g, offset);
+    uint32_t w[4] = {bits.x, bits.y, bits.z, bits.w};
+    V8 v;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      v.v[j * 2 + 0] =
+          __float2bfloat16(fmaf((float)(w[j] & 0xffffu) * (1.0f / 65536.0f), b, a));
+      v.v[j * 2 + 1] =
+          __float2bfloat16(fmaf((float)(w[j] >> 16) * (1.0f / 65536.0f), b, a));
+    }
+    *reinterpret_cast<V8*>(out + (uint64_t)g * 8) = v;
+  }
+}
+
 template <typename K>
 double bench(K kernel, __hip_bfloat16* buf, uint32_t n8, int blocks,
              int iters) {
@@ -138,13 +210,12 @@ int main() {
   __hip_bfloat16* buf;
   CHECK(hipMalloc(&buf, bytes));
 
-  for (int blocks : {2048, 4096, 8192}) {
+  for (int blocks : {4096, 8192, 16384}) {
     printf("blocks=%d\n", blocks);
-    printf("  strided EPT8 : %.2f TB/s\n", bench(rng_bf16<8>, buf, n8, blocks, 5));
-    printf("  strided EPT16: %.2f TB/s\n", bench(rng_bf16<16>, buf, n8, blocks, 5));
-    printf("  wide G2      : %.2f TB/s\n", bench(rng_bf16_wide<2>, buf, n8, blocks, 5));
-    printf("  wide G4      : %.2f TB/s\n", bench(rng_bf16_wide<4>, buf, n8, blocks, 5));
-    printf("  wide G8      : %.2f TB/s\n", bench(rng_bf16_wide<8>, buf, n8, blocks, 5));
+    printf("  uniform 2xphilox (v1): %.2f TB/s\n", bench(rng_bf16<8>, buf, n8, blocks, 5));
+    printf("  uniform 1xphilox (v2): %.2f TB/s\n", bench(uniform_bf16_u16, buf, n8, blocks, 5));
+    printf("  normal  2xphilox (v1): %.2f TB/s\n", bench(normal_bf16_u32, buf, n8, blocks, 5));
+    printf("  normal  1xphilox (v2): %.2f TB/s\n", bench(normal_bf16_u16, buf, n8, blocks, 5));
   }
   // memset reference ceiling
   hipEvent_t e0, e1;

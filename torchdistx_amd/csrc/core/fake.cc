@@ -22,6 +22,24 @@ namespace {
 
 constexpr c10::DispatchKey kFakeKey = c10::DispatchKey::Fake;
 
+// An index-less "cuda" device resolves to the current device at allocation
+// time in eager mode; fake tensors must report the same concrete index
+// (FSDP and friends compare devices exactly).
+c10::Device canonicalFakeDevice(c10::Device d) {
+  if (d.is_cuda() && !d.has_index()) {
+    const auto* guard =
+        c10::impl::device_guard_impl_registry[static_cast<size_t>(
+                                                  c10::DeviceType::CUDA)]
+            .load();
+    c10::DeviceIndex index = 0;
+    if (guard != nullptr && guard->deviceCount() > 0) {
+      index = guard->getDevice().index();
+    }
+    d = c10::Device{c10::DeviceType::CUDA, index};
+  }
+  return d;
+}
+
 c10::DispatchKeySet computeFakeKeySet(const at::Tensor& meta,
                                       c10::Device fake_device) {
   c10::DispatchKey backend =
@@ -36,11 +54,11 @@ c10::DispatchKeySet computeFakeKeySet(const at::Tensor& meta,
 }  // namespace
 
 FakeTensorImpl::FakeTensorImpl(at::Tensor meta, c10::Device fake_device)
-    : TensorImpl(computeFakeKeySet(meta, fake_device),
+    : TensorImpl(computeFakeKeySet(meta, canonicalFakeDevice(fake_device)),
                  meta.dtype(),
-                 fake_device),
+                 canonicalFakeDevice(fake_device)),
       meta_{std::move(meta)},
-      fake_device_{fake_device} {
+      fake_device_{canonicalFakeDevice(fake_device)} {
   TORCH_INTERNAL_ASSERT(meta_.is_meta(),
                         "FakeTensorImpl requires a meta shadow tensor.");
   set_storage_access_should_throw();
