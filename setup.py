@@ -57,6 +57,9 @@ setup(
         "torchdistx's capabilities"
     ),
     packages=[
+        "torchdistx",
+        "torchdistx.slowmo",
+        "torchdistx.optimizers",
         "torchdistx_amd",
         "torchdistx_amd.slowmo",
         "torchdistx_amd.optimizers",

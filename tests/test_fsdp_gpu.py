@@ -26,7 +26,11 @@ def nccl_world1():
     dist.destroy_process_group()
 
 
-def test_fsdp_with_deferred_init_param_init_fn(nccl_world1) -> None:
+@pytest.mark.parametrize("explicit_init_fn", [True, False])
+def test_fsdp_with_deferred_init(nccl_world1, explicit_init_fn) -> None:
+    # Without param_init_fn, FSDP's built-in torchdistx support (enabled by
+    # the `torchdistx` alias package) detects the fake parameters and calls
+    # materialize_module itself.
     from torch.distributed.fsdp import FullyShardedDataParallel as FSDP
 
     from torchdistx_amd import deferred_init, materialize_module
@@ -36,10 +40,13 @@ def test_fsdp_with_deferred_init_param_init_fn(nccl_world1) -> None:
     module = deferred_init(
         build_model, TINY, device="cuda", dtype=torch.float32
     )
+    kwargs = {}
+    if explicit_init_fn:
+        kwargs["param_init_fn"] = lambda m: materialize_module(m)
     fsdp = FSDP(
         module,
-        param_init_fn=lambda m: materialize_module(m),
         device_id=torch.cuda.current_device(),
+        **kwargs,
     )
     tokens = torch.randint(0, TINY.vocab_size, (2, 16), device="cuda")
     loss = fsdp(tokens[:, :-1]).float().mean()

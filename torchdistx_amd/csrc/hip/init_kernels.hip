@@ -37,10 +37,11 @@ namespace {
 
 constexpr int kBlock = 256;
 // Measured on gfx950 (scripts/rng_tune.hip): this ALU-heavy streaming
-// kernel keeps improving past the usual ~2048-block guideline; 8192 blocks
-// + 32-bit indexing reaches 3.2 TB/s on a 4 GiB bf16 fill (pure-store
-// ceiling: 6.35 TB/s via hipMemset).
-constexpr int kMaxBlocks = 8192;
+// kernel keeps improving past the usual ~2048-block guideline — at 16384
+// blocks the one-philox-per-store uniform kernel reaches 5.55 TB/s on a
+// 4 GiB bf16 fill (pure-store ceiling: 6.37 TB/s via hipMemset; normals
+// are transcendental-bound at ~2.7 TB/s).
+constexpr int kMaxBlocks = 16384;
 
 // ---------------------------------------------------------------------------
 // Philox4x32-10 (standard constants), producing 4 x uint32 per invocation.
