@@ -184,6 +184,69 @@ __global__ void uniform_bf16_u16(__hip_bfloat16* __restrict__ out,
   }
 }
 
+// ILP variant: two independent groups per iteration so the transcendental
+// chains of two Box-Muller batches interleave.
+__global__ void normal_bf16_u16x2(__hip_bfloat16* __restrict__ out,
+                                  uint32_t n8, float a, float b,
+                                  uint64_t seed, uint64_t offset) {
+  const uint32_t stride = gridDim.x * blockDim.x;
+  const uint32_t half = (n8 + 1) / 2;
+  for (uint32_t t = blockIdx.x * blockDim.x + threadIdx.x; t < half;
+       t += stride) {
+    uint32_t g0 = t;
+    uint32_t g1 = t + half;
+    uint4 b0 = philox10(seed, g0, offset);
+    uint4 b1 = philox10(seed, g1, offset);
+    uint32_t w0[4] = {b0.x, b0.y, b0.z, b0.w};
+    uint32_t w1[4] = {b1.x, b1.y, b1.z, b1.w};
+    V8 v0, v1;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float2 n0 = bm((float)(w0[j] & 0xffffu) * (1.0f / 65536.0f),
+                     (float)(w0[j] >> 16) * (1.0f / 65536.0f));
+      float2 n1 = bm((float)(w1[j] & 0xffffu) * (1.0f / 65536.0f),
+                     (float)(w1[j] >> 16) * (1.0f / 65536.0f));
+      v0.v[j * 2 + 0] = __float2bfloat16(fmaf(n0.x, b, a));
+      v0.v[j * 2 + 1] = __float2bfloat16(fmaf(n0.y, b, a));
+      v1.v[j * 2 + 0] = __float2bfloat16(fmaf(n1.x, b, a));
+      v1.v[j * 2 + 1] = __float2bfloat16(fmaf(n1.y, b, a));
+    }
+    *reinterpret_cast<V8*>(out + (uint64_t)g0 * 8) = v0;
+    if (g1 < n8) {
+      *reinterpret_cast<V8*>(out + (uint64_t)g1 * 8) = v1;
+    }
+  }
+}
+
+// Cheap-log variant: ln(u) for u = m * 2^-16 with integer m>=1 can be
+// computed from the float bit pattern: ln(u) = (log2(m) - 16) * ln2, and
+// log2(m) = exponent + log2(mantissa); v_log_f32 IS that — so instead try
+// replacing sincos with a single sin via angle-sum identity? Keep simple:
+// variant replacing __logf with __log2f * ln2 (same instr count, sanity).
+__global__ void normal_bf16_log2(__hip_bfloat16* __restrict__ out,
+                                 uint32_t n8, float a, float b,
+                                 uint64_t seed, uint64_t offset) {
+  const uint32_t stride = gridDim.x * blockDim.x;
+  for (uint32_t g = blockIdx.x * blockDim.x + threadIdx.x; g < n8;
+       g += stride) {
+    uint4 bits = philox10(seed, g, offset);
+    uint32_t w[4] = {bits.x, bits.y, bits.z, bits.w};
+    V8 v;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float u1 = fmaxf((float)(w[j] & 0xffffu) * (1.0f / 65536.0f),
+                       1.1754944e-38f);
+      float r = sqrtf(-1.3862943611f * __log2f(u1));
+      float sn, cs;
+      __sincosf(6.2831853071795865f * ((float)(w[j] >> 16) * (1.0f / 65536.0f)),
+                &sn, &cs);
+      v.v[j * 2 + 0] = __float2bfloat16(fmaf(r * cs, b, a));
+      v.v[j * 2 + 1] = __float2bfloat16(fmaf(r * sn, b, a));
+    }
+    *reinterpret_cast<V8*>(out + (uint64_t)g * 8) = v;
+  }
+}
+
 template <typename K>
 double bench(K kernel, __hip_bfloat16* buf, uint32_t n8, int blocks,
              int iters) {
@@ -216,6 +279,8 @@ int main() {
     printf("  uniform 1xphilox (v2): %.2f TB/s\n", bench(uniform_bf16_u16, buf, n8, blocks, 5));
     printf("  normal  2xphilox (v1): %.2f TB/s\n", bench(normal_bf16_u32, buf, n8, blocks, 5));
     printf("  normal  1xphilox (v2): %.2f TB/s\n", bench(normal_bf16_u16, buf, n8, blocks, 5));
+    printf("  normal  v2 ILPx2     : %.2f TB/s\n", bench(normal_bf16_u16x2, buf, n8, blocks, 5));
+    printf("  normal  v2 log2      : %.2f TB/s\n", bench(normal_bf16_log2, buf, n8, blocks, 5));
   }
   // memset reference ceiling
   hipEvent_t e0, e1;
