@@ -92,11 +92,14 @@ __device__ __forceinline__ float u16_to_uniform(uint32_t x) {
   return static_cast<float>(x & 0xffffu) * (1.0f / 65536.0f);
 }
 
-// Two uniforms -> two standard normals (Box-Muller).
+// Two uniforms -> two standard normals (Box-Muller). -2*ln(u) is computed
+// as -2*ln2*log2(u): __log2f maps straight to v_log_f32, measurably
+// faster than __logf's wrapper (2.69 -> 3.15 TB/s on the bf16 normal
+// kernel, scripts/rng_tune.hip).
 __device__ __forceinline__ float2 box_muller(float u1, float u2) {
-  // Guard u1 away from 0 so log() stays finite.
+  // Guard u1 away from 0 so log stays finite.
   u1 = fmaxf(u1, 1.1754944e-38f);
-  float r = sqrtf(-2.0f * __logf(u1));
+  float r = sqrtf(-1.3862943611198906f * __log2f(u1));
   float s, c;
   __sincosf(6.2831853071795865f * u2, &s, &c);
   return make_float2(r * c, r * s);
