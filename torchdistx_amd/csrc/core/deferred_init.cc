@@ -163,6 +163,13 @@ struct OpNode {
 
 std::atomic<uint64_t> next_op_nr{1};
 
+// Serializes tape mutation (recordOp) and tape replay (materialize /
+// recordInfo): materialization runs with the GIL released, so without
+// this a second thread could record into or replay from the same node
+// graph mid-walk. Recursive because a terminal op (aten::item) can
+// materialize while a recording is on the stack.
+std::recursive_mutex tape_mutex;
+
 std::shared_ptr<TensorRecord> getRecord(FakeTensorImpl* fake) {
   return std::static_pointer_cast<TensorRecord>(fake->getData(kDeferredKey));
 }
@@ -200,6 +207,7 @@ void recordOp(std::string name,
               at::ThreadLocalState tls,
               torch::jit::Stack& stack,
               size_t rets_begin) {
+  std::lock_guard<std::recursive_mutex> lock{tape_mutex};
   auto node = std::make_shared<OpNode>();
   node->op_nr = next_op_nr.fetch_add(1, std::memory_order_relaxed);
 
@@ -597,15 +605,6 @@ bool outputsAlias(const OpNode& node, const c10::Storage& storage) {
   return false;
 }
 
-bool sharesOutputStorage(const OpNode& a, const OpNode& b) {
-  for (const c10::Storage& s : a.output_storages) {
-    if (s && outputsAlias(b, s)) {
-      return true;
-    }
-  }
-  return false;
-}
-
 // The ordered replay set for the tensor behind `rec` (fills the role of
 // the reference's buildCallStack + getLastInPlaceOpNode,
 // deferred_init.cc:506-622, but is driven by the alias group instead of a
@@ -796,6 +795,7 @@ bool canMaterialize(const at::Tensor& tensor) noexcept {
 }
 
 std::optional<RecordInfo> recordInfo(const at::Tensor& tensor) {
+  std::lock_guard<std::recursive_mutex> lock{tape_mutex};
   auto* fake = asFake(tensor);
   if (fake == nullptr) {
     return std::nullopt;
@@ -832,6 +832,7 @@ at::Tensor materializeTensor(const at::Tensor& tensor) {
                     "`tensor` is fake but carries no deferred-init record, "
                     "so it cannot be materialized.");
 
+  std::lock_guard<std::recursive_mutex> lock{tape_mutex};
   auto node = rec->desc.node;
   for (const auto& n : buildCallStack(*rec)) {
     replayNode(n);
