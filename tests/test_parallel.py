@@ -108,3 +108,37 @@ def test_default_dtype_is_pinned_at_record_time() -> None:
     assert m.tok_emb.weight.dtype == torch.bfloat16
     # Buffers created with an explicit fp32 dtype stay fp32.
     assert m.rope_cos.dtype == torch.float32
+
+
+def _expert_shard_worker(rank, world):
+    from torchdistx_amd import deferred_init, _C
+    from torchdistx_amd.models import TINY_MOE, build_model
+    from torchdistx_amd.parallel import materialize_experts_sharded
+
+    torch.manual_seed(3)
+    m = deferred_init(build_model, TINY_MOE, device="cpu", dtype=torch.float32)
+    owners = materialize_experts_sharded(m)
+    state = {}
+    for name, p in m.named_parameters():
+        state[name] = None if _C.is_fake(p) else float(p.detach().double().sum())
+    return owners, state
+
+
+def test_expert_sharded_materialization() -> None:
+    results = run_distributed(_expert_shard_worker, 2)
+    owners0, s0 = results[0]
+    owners1, s1 = results[1]
+    assert owners0 == owners1
+    assert set(owners0.values()) == {0, 1}
+
+    for name in s0:
+        is_expert = ".experts." in name
+        if not is_expert:
+            # Shared parameters materialize on every rank (identically up
+            # to the partition-invariant GPU path; on CPU both replay the
+            # full shared subgraph from the same seed).
+            assert s0[name] is not None and s1[name] is not None, name
+        else:
+            # Every expert parameter is real on exactly one rank.
+            real_on = [r for r, s in enumerate((s0, s1)) if s[name] is not None]
+            assert len(real_on) == 1, name

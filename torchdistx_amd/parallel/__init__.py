@@ -1,4 +1,5 @@
 from torchdistx_amd.parallel.sharded_materialize import (  # noqa: F401
     assign_owners,
+    materialize_experts_sharded,
     materialize_module_distributed,
 )

@@ -28,7 +28,7 @@
 #     trusted to share RNG state (e.g. mixed seeds) but must end
 #     bitwise-identical.
 
-from typing import Callable, Dict, List, Optional, Tuple
+from typing import Callable, Dict, List, Optional, Tuple, Union
 
 import torch
 import torch.distributed as dist
@@ -130,3 +130,44 @@ def materialize_module_distributed(
     for h in handles:
         h.wait()
     return {i: owners[i] for i in range(len(entries))}
+
+
+def materialize_experts_sharded(
+    module: Module,
+    expert_fqn_fragment: str = "experts",
+    process_group: Optional[dist.ProcessGroup] = None,
+) -> Dict[str, int]:
+    """Expert-sharded materialization for MoE models (the Mixtral-style
+    layout in torchdistx_amd.models): expert submodules are assigned
+    round-robin to ranks and only the owner materializes them (expert
+    parallelism — each rank holds its experts and nothing else), while all
+    non-expert ("shared") parameters are materialized on every rank
+    bitwise-identically via the partition-invariant Philox tape. No
+    communication at all: ownership is decided by deterministic rank
+    arithmetic, values by the pinned counters.
+
+    Returns {expert module fqn -> owner rank}.
+    """
+    group = process_group or (dist.group.WORLD if dist.is_initialized() else None)
+    world = dist.get_world_size(group) if group is not None else 1
+    rank = dist.get_rank(group) if group is not None else 0
+
+    # Deterministic expert enumeration by fully-qualified name.
+    expert_fqns = [
+        name
+        for name, sub in module.named_modules()
+        if f".{expert_fqn_fragment}." in f".{name}."
+        and "." not in name.split(f"{expert_fqn_fragment}.")[-1]
+    ]
+    owners = {fqn: i % world for i, fqn in enumerate(expert_fqns)}
+
+    all_modules = dict(module.named_modules())
+    skip_ids = set()
+    for fqn, owner in owners.items():
+        if owner != rank:
+            skip_ids.update(id(m) for m in all_modules[fqn].modules())
+
+    from torchdistx_amd.deferred_init import materialize_module
+
+    materialize_module(module, check_fn=lambda sub: id(sub) not in skip_ids)
+    return owners
