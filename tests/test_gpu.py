@@ -198,3 +198,33 @@ def test_gpt2_xl_materialize_bf16() -> None:
     assert n == GPT2_XL.n_params
     del m
     torch.cuda.empty_cache()
+
+
+def test_tdx_cast_copy() -> None:
+    src = torch.randn(1000003, device="cuda", dtype=torch.float32)
+    dst = torch.empty(1000003, device="cuda", dtype=torch.bfloat16)
+    torch.ops.tdx.copy_(dst, src)
+    assert torch.equal(dst, src.to(torch.bfloat16))
+    # same-dtype memcpy path
+    dst2 = torch.empty_like(src)
+    torch.ops.tdx.copy_(dst2, src)
+    assert torch.equal(dst2, src)
+
+
+def test_deferred_to_dtype_copy_replay() -> None:
+    # A .to(dtype) inside deferred init records copy ops; replay must run
+    # them through the cast-copy kernel and match eager numerics.
+    from torchdistx_amd import deferred_init, materialize_tensor
+    from torch.nn import Module, Parameter
+
+    class M(Module):
+        def __init__(self):
+            super().__init__()
+            w = torch.randn(64, 64, device="cuda", dtype=torch.float32)
+            self.p = Parameter(w.to(torch.bfloat16))
+
+    torch.manual_seed(11)
+    m = deferred_init(M)
+    p = materialize_tensor(m.p)
+    assert p.dtype == torch.bfloat16 and p.is_cuda
+    assert p.detach().float().abs().sum().item() > 0

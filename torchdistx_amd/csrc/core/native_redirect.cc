@@ -21,6 +21,7 @@ constexpr Redirect kRedirects[] = {
     {"aten::normal_", "", "normal_", true},
     {"aten::fill_", "Scalar", "fill_", true},
     {"aten::zero_", "", "zero_", false},
+    {"aten::copy_", "", "copy_", true},
 };
 
 bool requireNative() {
@@ -69,6 +70,19 @@ bool tryNativeInitRedirect(
   if (redirect->needs_float_dtype) {
     auto st = self.scalar_type();
     if (st != at::kFloat && st != at::kBFloat16 && st != at::kHalf) {
+      return false;
+    }
+  }
+  if (std::strcmp(redirect->tdx_name, "copy_") == 0) {
+    // tdx::copy_ covers the same-device contiguous same-shape case only.
+    if (stack.size() < 2 || !stack[1].isTensor()) {
+      return false;
+    }
+    const at::Tensor& src = stack[1].toTensor();
+    auto sst = src.scalar_type();
+    if (!src.defined() || !src.is_cuda() || !src.is_contiguous() ||
+        !src.sizes().equals(self.sizes()) ||
+        (sst != at::kFloat && sst != at::kBFloat16 && sst != at::kHalf)) {
       return false;
     }
   }
