@@ -265,3 +265,34 @@ def test_deferred_init_forward_shapes_on_fake() -> None:
         y = m(x)
     assert is_fake(y)
     assert y.shape == (5, 3)
+
+
+def test_lazy_modules_with_dry_run() -> None:
+    # Reference contract (docs/src/deferred_init.rst:152-170): lazy modules
+    # work when the dry-run happens inside the deferred function.
+    def my_lazy(out_features):
+        m = torch.nn.LazyLinear(out_features)
+        m(torch.ones([10, 10]))
+        return m
+
+    m = deferred_init(my_lazy, 10)
+    assert is_deferred(m)
+    assert m.weight.shape == (10, 10)
+    materialize_module(m)
+    assert m(torch.ones(2, 10)).shape == (2, 10)
+
+
+def test_inference_tensor_rejected_at_materialization() -> None:
+    # Reference contract (docs/src/deferred_init.rst:200-202): inference
+    # tensors cannot participate; the error comes at materialization.
+    with torch.inference_mode():
+        inf = torch.tensor(2.0)
+
+    class M(Module):
+        def __init__(self):
+            super().__init__()
+            self.p = Parameter(torch.zeros(3) + inf)
+
+    m = deferred_init(M)
+    with pytest.raises(RuntimeError, match="[Ii]nference"):
+        materialize_tensor(m.p)

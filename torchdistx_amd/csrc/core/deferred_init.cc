@@ -149,6 +149,9 @@ struct InputSlot {
   // after the other family's fakes are gone. Acyclic by construction:
   // every held node has a smaller op_nr.
   std::vector<std::shared_ptr<OpNode>> history;
+  // Inference tensors have no version counter, so their mutations cannot
+  // be tracked; replay rejects them with a clear error.
+  bool inference = false;
 };
 
 struct OpNode {
@@ -235,7 +238,11 @@ void recordOp(std::string name,
                    input_storages.push_back(fake->meta_tensor().storage());
                    return at::Tensor{};
                  }
-                 if (t.defined()) {
+                 if (t.defined() && t.is_inference()) {
+                   InputSlot slot;
+                   slot.inference = true;
+                   node->input_slots.push_back(std::move(slot));
+                 } else if (t.defined()) {
                    node->input_slots.push_back(
                        InputSlot{{}, tensorVersion(t), {}});
                  } else {
@@ -714,12 +721,12 @@ torch::jit::Stack materializeArguments(OpNode& node) {
                             "not yet materialized");
       return dep.node->outputs.at(dep.index);
     }
+    TORCH_CHECK(!slot.inference,
+                "An inference tensor was used while recording `",
+                node.op->name,
+                "`; in-place updates to inference tensors cannot be "
+                "tracked, so the tape cannot be replayed faithfully.");
     if (t.defined() && slot.external_version.has_value()) {
-      TORCH_CHECK(!t.is_inference(),
-                  "An inference tensor was used while recording `",
-                  node.op->name,
-                  "`; inference tensors cannot participate in deferred "
-                  "initialization.");
       TORCH_CHECK(tensorVersion(t) == *slot.external_version,
                   "The external tensor used by the recorded op `",
                   node.op->name,
