@@ -19,7 +19,11 @@ def nccl_world1():
     import torch.distributed as dist
 
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-    os.environ.setdefault("MASTER_PORT", "29551")
+    import socket
+
+    with socket.socket() as _s:
+        _s.bind(("127.0.0.1", 0))
+        os.environ["MASTER_PORT"] = str(_s.getsockname()[1])
     torch.cuda.set_device(0)
     dist.init_process_group("nccl", rank=0, world_size=1)
     yield dist

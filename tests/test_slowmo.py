@@ -17,7 +17,11 @@ from torchdistx_amd.slowmo import SlowMomentumOptimizer
 @pytest.fixture()
 def single_proc_group():
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-    os.environ.setdefault("MASTER_PORT", "29511")
+    import socket
+
+    with socket.socket() as _s:
+        _s.bind(("127.0.0.1", 0))
+        os.environ["MASTER_PORT"] = str(_s.getsockname()[1])
     dist.init_process_group("gloo", rank=0, world_size=1)
     yield
     dist.destroy_process_group()
