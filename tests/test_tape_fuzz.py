@@ -28,19 +28,27 @@ def _build_program(rng: random.Random, n_steps: int):
             n_tensors += 1
             continue
         r = rng.random()
-        if r < 0.25:
+        if r < 0.18:
             ops.append(("view", any_tensor()))
             n_tensors += 1
-        elif r < 0.4:
+        elif r < 0.30:
             ops.append(("slice", any_tensor(), rng.randrange(2)))
             n_tensors += 1
-        elif r < 0.6:
+        elif r < 0.38:
+            ops.append(("transpose", any_tensor()))
+            n_tensors += 1
+        elif r < 0.44:
+            ops.append(("select", any_tensor(), rng.randrange(2)))
+            n_tensors += 1
+        elif r < 0.60:
             ops.append(("add_", any_tensor(), rng.uniform(-1, 1)))
-        elif r < 0.7:
+        elif r < 0.70:
             ops.append(("mul_", any_tensor(), rng.uniform(0.5, 1.5)))
-        elif r < 0.8:
+        elif r < 0.80:
             ops.append(("fill_", any_tensor(), rng.uniform(-3, 3)))
-        elif r < 0.9:
+        elif r < 0.86:
+            ops.append(("iota_", any_tensor()))
+        elif r < 0.93:
             ops.append(("copy_", any_tensor(), any_tensor()))
         else:
             ops.append(("addt", any_tensor(), any_tensor()))
@@ -50,17 +58,50 @@ def _build_program(rng: random.Random, n_steps: int):
 
 def _run_program(ops):
     ts = []
+    fam = []  # structural alias family per tensor (identical for fake/real)
     for op in ops:
         kind = op[0]
         if kind == "new":
             _, n, val = op
             ts.append(torch.full([n, n], val))
+            fam.append(len(fam))
         elif kind == "view":
-            ts.append(ts[op[1]].view(-1))
+            base = ts[op[1]]
+            # view() demands contiguity; reshape handles every case (and
+            # copies for non-contiguous inputs, identically in both runs).
+            ts.append(base.reshape(-1) if not base.is_contiguous() else base.view(-1))
+            # reshape of a non-contiguous tensor copies (fresh family)
+            fam.append(len(fam) if not base.is_contiguous() else fam[op[1]])
         elif kind == "slice":
             base = ts[op[1]]
-            half = base.shape[0] // 2
-            ts.append(base[half:] if op[2] else base[:half])
+            if base.dim() == 0:
+                ts.append(base.reshape(1))
+            else:
+                half = max(1, base.shape[0] // 2)
+                ts.append(base[half:] if op[2] else base[:half])
+            fam.append(fam[op[1]])
+        elif kind == "transpose":
+            base = ts[op[1]]
+            if base.dim() == 2:
+                ts.append(base.t())
+                fam.append(fam[op[1]])
+            else:
+                ts.append(base.reshape(-1))
+                fam.append(fam[op[1]] if base.is_contiguous() else len(fam))
+        elif kind == "select":
+            base = ts[op[1]]
+            if base.dim() == 0:
+                ts.append(base.reshape(1))
+            else:
+                ts.append(base[min(op[2], base.shape[0] - 1)])
+            fam.append(fam[op[1]])
+        elif kind == "iota_":
+            t = ts[op[1]]
+            # non-trivial in-place chain: arange copied into (possibly a
+            # non-contiguous view of) t
+            t.copy_(
+                torch.arange(t.numel(), dtype=t.dtype).reshape(t.shape)
+            )
         elif kind == "add_":
             ts[op[1]].add_(op[2])
         elif kind == "mul_":
@@ -69,7 +110,9 @@ def _run_program(ops):
             ts[op[1]].fill_(op[2])
         elif kind == "copy_":
             dst, src = ts[op[1]], ts[op[2]]
-            if dst.numel() == src.numel():
+            # skip aliased pairs: torch rejects internally-overlapping
+            # copies, and the structural check is fake/real-identical
+            if dst.numel() == src.numel() and fam[op[1]] != fam[op[2]]:
                 dst.copy_(src.reshape(dst.shape))
         elif kind == "addt":
             a, b = ts[op[1]], ts[op[2]]
@@ -77,6 +120,7 @@ def _run_program(ops):
                 ts.append(a + b.reshape(a.shape))
             else:
                 ts.append(a + 1)
+            fam.append(len(fam))
     return ts
 
 
