@@ -296,3 +296,18 @@ def test_inference_tensor_rejected_at_materialization() -> None:
     m = deferred_init(M)
     with pytest.raises(RuntimeError, match="[Ii]nference"):
         materialize_tensor(m.p)
+
+
+def test_beyond_hardware_scale_instantiation() -> None:
+    # The reference's qualitative capability (fake_tensor.rst:65) at scale:
+    # a 405B-parameter model (812 GB bf16) is fully inspectable on this
+    # GPU-less CI machine with ~zero memory.
+    from torchdistx_amd.models import LLAMA3_405B, build_model
+
+    m = deferred_init(build_model, LLAMA3_405B, device="cpu",
+                      dtype=torch.bfloat16)
+    assert is_deferred(m)
+    n = sum(p.numel() for p in m.parameters())
+    assert n == LLAMA3_405B.n_params
+    assert n > 400e9
+    assert m.blocks[0].attn.wq.weight.shape == (16384, 16384)

@@ -171,3 +171,22 @@ def test_gossip_hierarchical_two_nodes() -> None:
     results = run_distributed(_gossip_two_gpu_nodes, 4)
     for got in results:
         assert got == pytest.approx([1.5] * 4)
+
+
+def _topology_determinism(rank, world):
+    state = GossipGraDState(
+        num_modules=1,
+        local_process_group=dist.new_group([rank]),
+        num_nodes=world,
+        master_process_group=dist.group.WORLD,
+        proc_per_node=1,
+        random_seed=777,
+    )
+    return [list(next(state.topologies)) for _ in range(4)]
+
+
+def test_topology_sequence_identical_across_workers() -> None:
+    # Every worker must generate the same rotation sequence from the seed,
+    # or peers would disagree mid-training.
+    results = run_distributed(_topology_determinism, 2)
+    assert results[0] == results[1]

@@ -1,4 +1,5 @@
 from torchdistx_amd.models.configs import (  # noqa: F401
+    LLAMA3_405B,
     CONFIGS,
     GPT2_XL,
     LLAMA3_8B,

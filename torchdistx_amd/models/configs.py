@@ -131,6 +131,21 @@ LLAMA3_70B = TransformerConfig(
     max_seq_len=8192,
 )
 
+# Too large for ANY single device (812 GB bf16): exists to exercise the
+# fake/deferred layer at beyond-hardware scale (the reference's qualitative
+# "model too big for the machine" capability, docs/src/fake_tensor.rst:65)
+# and the sharded materialization planners.
+LLAMA3_405B = TransformerConfig(
+    name="llama3-405b",
+    vocab_size=128256,
+    dim=16384,
+    n_layers=126,
+    n_heads=128,
+    n_kv_heads=8,
+    ffn_hidden=53248,
+    max_seq_len=8192,
+)
+
 MIXTRAL_8X22B = TransformerConfig(
     name="mixtral-8x22b",
     vocab_size=32768,
@@ -153,6 +168,7 @@ CONFIGS = {
         GPT2_XL,
         LLAMA3_8B,
         LLAMA3_70B,
+        LLAMA3_405B,
         MIXTRAL_8X22B,
     ]
 }
