@@ -56,14 +56,14 @@ def _build_program(rng: random.Random, n_steps: int):
     return ops
 
 
-def _run_program(ops):
+def _run_program(ops, device="cpu"):
     ts = []
     fam = []  # structural alias family per tensor (identical for fake/real)
     for op in ops:
         kind = op[0]
         if kind == "new":
             _, n, val = op
-            ts.append(torch.full([n, n], val))
+            ts.append(torch.full([n, n], val, device=device))
             fam.append(len(fam))
         elif kind == "view":
             base = ts[op[1]]
@@ -100,7 +100,9 @@ def _run_program(ops):
             # non-trivial in-place chain: arange copied into (possibly a
             # non-contiguous view of) t
             t.copy_(
-                torch.arange(t.numel(), dtype=t.dtype).reshape(t.shape)
+                torch.arange(
+                    t.numel(), dtype=t.dtype, device=t.device
+                ).reshape(t.shape)
             )
         elif kind == "add_":
             ts[op[1]].add_(op[2])
@@ -124,17 +126,16 @@ def _run_program(ops):
     return ts
 
 
-@pytest.mark.parametrize("seed", range(30))
-def test_random_program_replay_matches_eager(seed) -> None:
+def _check_seed(seed, device):
     rng = random.Random(seed)
     ops = _build_program(rng, n_steps=25)
 
-    eager = _run_program(ops)
+    eager = _run_program(ops, device)
 
     class Holder(Module):
         def __init__(self):
             super().__init__()
-            self.tensors = _run_program(ops)
+            self.tensors = _run_program(ops, device)
 
     holder = deferred_init(Holder)
 
@@ -143,3 +144,19 @@ def test_random_program_replay_matches_eager(seed) -> None:
     for i in order:
         got = materialize_tensor(holder.tensors[i])
         assert torch.equal(got, eager[i]), (seed, i, ops)
+
+
+@pytest.mark.parametrize("seed", range(30))
+def test_random_program_replay_matches_eager(seed) -> None:
+    _check_seed(seed, "cpu")
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("seed", range(30))
+def test_random_program_replay_matches_eager_gpu_native(seed) -> None:
+    # All fuzz ops are deterministic, so the GPU run differentially
+    # validates the native tdx kernel redirect (fill_/zero_/copy_) against
+    # stock eager execution under arbitrary aliasing.
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    _check_seed(seed + 10_000, "cuda")
