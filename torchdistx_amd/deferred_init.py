@@ -17,7 +17,7 @@
 
 from typing import Callable, Dict, Optional, TypeVar, Union
 
-import torch
+
 from torch import Tensor
 from torch.nn import Module, Parameter
 from torch.utils.weak import WeakTensorKeyDictionary

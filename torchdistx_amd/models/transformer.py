@@ -9,7 +9,6 @@
 # deferred-init tape records (empty / uniform_ / normal_ / zero_ / fill_ /
 # copy_), see SURVEY.md section 2.7.
 
-import math
 from typing import Optional
 
 import torch

@@ -28,11 +28,11 @@
 #     trusted to share RNG state (e.g. mixed seeds) but must end
 #     bitwise-identical.
 
-from typing import Callable, Dict, List, Optional, Tuple, Union
+from typing import Callable, Dict, List, Optional, Tuple
 
 import torch
 import torch.distributed as dist
-from torch.nn import Module, Parameter
+from torch.nn import Module
 
 from torchdistx_amd import _C
 from torchdistx_amd.deferred_init import _restore_class
