@@ -62,7 +62,8 @@ def main():
     )
 
     for step in range(10):
-        tokens = torch.randint(0, cfg.vocab_size, (4, 256), device="cuda")
+        seq = min(256, cfg.max_seq_len)
+        tokens = torch.randint(0, cfg.vocab_size, (4, seq), device="cuda")
         optim.zero_grad()
         loss = fsdp(tokens[:, :-1]).float().mean()
         loss.backward()

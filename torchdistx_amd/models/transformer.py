@@ -75,6 +75,11 @@ def _rope_cache(head_dim: int, max_seq_len: int, base: float = 500000.0):
 def _apply_rope(x: Tensor, cos: Tensor, sin: Tensor) -> Tensor:
     # x: [B, H, S, D]
     s = x.shape[-2]
+    if s > cos.shape[0]:
+        raise ValueError(
+            f"sequence length {s} exceeds the RoPE cache "
+            f"(max_seq_len={cos.shape[0]})"
+        )
     cos = cos[:s].to(x.dtype)
     sin = sin[:s].to(x.dtype)
     x1, x2 = x.chunk(2, dim=-1)
