@@ -65,7 +65,11 @@ def main():
         seq = min(256, cfg.max_seq_len)
         tokens = torch.randint(0, cfg.vocab_size, (4, seq), device="cuda")
         optim.zero_grad()
-        loss = fsdp(tokens[:, :-1]).float().mean()
+        logits = fsdp(tokens[:, :-1])
+        loss = torch.nn.functional.cross_entropy(
+            logits.reshape(-1, logits.shape[-1]).float(),
+            tokens[:, 1:].reshape(-1),
+        )
         loss.backward()
         optim.step()
         if dist.get_rank() == 0:
