@@ -26,6 +26,7 @@ ext_modules = [
             os.path.join(CSRC, "core", "fake.cc"),
             os.path.join(CSRC, "core", "deferred_init.cc"),
             os.path.join(CSRC, "core", "native_redirect.cc"),
+            os.path.join(CSRC, "core", "tdx_ops.cc"),
             os.path.join(CSRC, "core", "module.cc"),
         ],
         extra_compile_args=["-O2", "-std=c++17", "-fvisibility=hidden"],

@@ -79,6 +79,9 @@ struct RecordedOp {
   std::function<void(torch::jit::Stack&)> run;
   std::vector<c10::IValue> args;  // deep-copied; fake slots -> undefined
   at::ThreadLocalState tls;       // captured with DeferredInit excluded
+  // Pinned Philox (seed, counter offset) for RNG ops; also consumed by the
+  // slice-materialization fast path (materializeTensorShard).
+  std::optional<std::pair<uint64_t, uint64_t>> philox;
 };
 
 // Pinned Philox state for recorded RNG ops: derived from the default
@@ -89,52 +92,63 @@ struct RecordedOp {
 // the native CDNA4 kernels run the replay).
 constexpr uint64_t kPhiloxStridePerOp = 4;
 
-// Session state: the seed is a 64-bit nonce drawn from the default CPU
-// generator when the outermost deferred_init is entered, and slots count
-// from 1 within the session. Consequences (all intentional):
-//   * torch.manual_seed(S) before deferred_init fixes the init bits;
-//   * two deferred_init sessions without re-seeding produce different
-//     inits (the second nonce differs), matching eager expectations;
+// Session state: the seed is a 64-bit nonce derived (by hashing, WITHOUT
+// consuming) from the default CPU generator's state when the outermost
+// deferred_init is entered, and slots count from 1 within the session.
+// Consequences (all intentional):
+//   * the pinned init bits are a pure function of the RNG state at
+//     deferred_init entry — torch.manual_seed(S) fixes them, exactly like
+//     eager init is a function of the generator state at construction;
+//   * the generator stream is never perturbed, so CPU replay (which uses
+//     the stock generator) stays bitwise-equal to eager construction;
 //   * within one tape, replaying any subset on any rank gives identical
 //     bits (partition-invariant sharded materialization).
 std::atomic<uint64_t> session_rng_slot{1};
 std::atomic<uint64_t> session_rng_seed{0};
-std::atomic<bool> session_nonce_drawn{false};
 
 bool isRngOpName(const std::string& name) {
   return name == "aten::uniform_" || name == "aten::normal_";
 }
 
 void beginRngSession() {
-  // Lazy: the nonce is drawn from the CPU generator only when the first
-  // GPU-targeted RNG op is recorded, so pure-CPU tapes never perturb the
-  // generator stream and their replay stays bitwise-equal to eager init.
-  session_nonce_drawn.store(false, std::memory_order_relaxed);
+  // FNV-1a over the seed and a prefix of the generator state: equal state
+  // -> equal nonce; any consumption or re-seeding -> different nonce.
+  auto gen = at::globalContext().defaultGenerator(c10::DeviceType::CPU);
+  uint64_t h = 1469598103934665603ull;
+  auto mix = [&h](uint64_t v) {
+    for (int i = 0; i < 8; ++i) {
+      h = (h ^ ((v >> (8 * i)) & 0xff)) * 1099511628211ull;
+    }
+  };
+  {
+    std::lock_guard<std::mutex> lock(gen.mutex());
+    mix(gen.current_seed());
+    at::Tensor state = gen.get_state();
+    const auto* bytes = state.const_data_ptr<uint8_t>();
+    int64_t take = std::min<int64_t>(state.numel(), 64);
+    uint64_t word = 0;
+    for (int64_t i = 0; i < take; ++i) {
+      word = (word << 8) | bytes[i];
+      if ((i & 7) == 7) {
+        mix(word);
+        word = 0;
+      }
+    }
+  }
+  session_rng_seed.store(h, std::memory_order_relaxed);
   session_rng_slot.store(1, std::memory_order_relaxed);
 }
 
-uint64_t sessionNonce() {
-  if (!session_nonce_drawn.exchange(true, std::memory_order_relaxed)) {
-    auto gen = at::globalContext().defaultGenerator(c10::DeviceType::CPU);
-    uint64_t nonce;
-    {
-      std::lock_guard<std::mutex> lock(gen.mutex());
-      nonce = at::check_generator<at::CPUGeneratorImpl>(gen)->random64();
-    }
-    session_rng_seed.store(nonce, std::memory_order_relaxed);
-  }
-  return session_rng_seed.load(std::memory_order_relaxed);
-}
-
-// Pins (seed, counter-offset) for an RNG op whose target lives on the GPU
-// (where the tdx Philox kernels consume it). CPU-targeted RNG ops replay
-// through the stock generator to preserve eager bit-parity.
+// Pins (seed, counter-offset) for every recorded RNG op. The pins drive
+// the tdx Philox kernels on GPU replay and the slice-materialization fast
+// path on any device; plain CPU replay ignores them (it uses the stock
+// generator for eager bit-parity).
 std::optional<std::pair<uint64_t, uint64_t>> pinPhiloxForOp(
-    const std::string& name, bool cuda_target) {
-  if (!cuda_target || !isRngOpName(name)) {
+    const std::string& name) {
+  if (!isRngOpName(name)) {
     return std::nullopt;
   }
-  uint64_t seed = sessionNonce();
+  uint64_t seed = session_rng_seed.load(std::memory_order_relaxed);
   uint64_t slot = session_rng_slot.fetch_add(1, std::memory_order_relaxed);
   return std::make_pair(seed, slot * kPhiloxStridePerOp);
 }
@@ -209,7 +223,9 @@ void recordOp(std::string name,
               std::vector<c10::IValue> args,
               at::ThreadLocalState tls,
               torch::jit::Stack& stack,
-              size_t rets_begin) {
+              size_t rets_begin,
+              std::optional<std::pair<uint64_t, uint64_t>> philox =
+                  std::nullopt) {
   std::lock_guard<std::recursive_mutex> lock{tape_mutex};
   auto node = std::make_shared<OpNode>();
   node->op_nr = next_op_nr.fetch_add(1, std::memory_order_relaxed);
@@ -254,7 +270,7 @@ void recordOp(std::string name,
   }
 
   node->op = RecordedOp{std::move(name), std::move(run), std::move(args),
-                        std::move(tls)};
+                        std::move(tls), philox};
 
   // Output pass: stamp (or restamp) each fake output's record.
   visitTensors(stack, rets_begin, stack.size(), [&](const at::Tensor& t) {
@@ -398,13 +414,7 @@ void deferredInitHandler(const c10::OperatorHandle& op,
   });
 
   if (has_fake_arg || has_fake_ret) {
-    bool cuda_target = false;
-    visitTensors(*stack, rets_begin, stack->size(), [&](const at::Tensor& t) {
-      if (auto* fake = asFake(t)) {
-        cuda_target = cuda_target || fake->fake_device().is_cuda();
-      }
-    });
-    auto philox = pinPhiloxForOp(schema.operator_name().name, cuda_target);
+    auto philox = pinPhiloxForOp(schema.operator_name().name);
     recordOp(
         schema.operator_name().name,
         [handle = op, philox](torch::jit::Stack& s) {
@@ -417,7 +427,7 @@ void deferredInitHandler(const c10::OperatorHandle& op,
           }
           handle.callBoxed(s);
         },
-        std::move(saved), std::move(tls), *stack, rets_begin);
+        std::move(saved), std::move(tls), *stack, rets_begin, philox);
   }
 }
 
@@ -827,6 +837,168 @@ std::optional<RecordInfo> recordInfo(const at::Tensor& tensor) {
     }
   }
   return info;
+}
+
+namespace {
+
+bool chainNameIs(const std::string& name, const char* prefix) {
+  return name.rfind(prefix, 0) == 0;
+}
+
+// One value-affecting step of a simple init chain.
+struct ChainStep {
+  enum class Kind { kFactory, kUniform, kNormal, kFill, kZero, kPass } kind;
+  double p0 = 0.0;  // from / mean / fill value
+  double p1 = 0.0;  // to / std
+  std::optional<std::pair<uint64_t, uint64_t>> philox;
+};
+
+ChainStep classifyChainNode(const OpNode& node) {
+  TORCH_CHECK(node.op.has_value(),
+              "slice materialization: the tape segment for `",
+              (node.materialized ? "an already-materialized op"
+                                 : "a freed op"),
+              "` is gone; materialize the tensor fully instead.");
+  const RecordedOp& op = *node.op;
+  ChainStep step;
+  auto scalarArg = [&](size_t i) -> double {
+    const c10::IValue& v = op.args.at(i);
+    if (v.isDouble()) {
+      return v.toDouble();
+    }
+    if (v.isInt()) {
+      return static_cast<double>(v.toInt());
+    }
+    if (v.isScalar()) {
+      return v.toScalar().toDouble();
+    }
+    TORCH_CHECK(false, "slice materialization: unexpected argument type `",
+                v.tagKind(), "` in `", op.name, "`");
+  };
+  if (chainNameIs(op.name, "aten::empty")) {
+    step.kind = ChainStep::Kind::kFactory;
+  } else if (chainNameIs(op.name, "aten::zeros")) {
+    step.kind = ChainStep::Kind::kZero;
+  } else if (chainNameIs(op.name, "aten::ones")) {
+    step.kind = ChainStep::Kind::kFill;
+    step.p0 = 1.0;
+  } else if (chainNameIs(op.name, "aten::full")) {
+    step.kind = ChainStep::Kind::kFill;
+    step.p0 = scalarArg(1);
+  } else if (op.name == "aten::uniform_") {
+    step.kind = ChainStep::Kind::kUniform;
+    step.p0 = scalarArg(1);
+    step.p1 = scalarArg(2);
+    step.philox = op.philox;
+  } else if (op.name == "aten::normal_") {
+    step.kind = ChainStep::Kind::kNormal;
+    step.p0 = scalarArg(1);
+    step.p1 = scalarArg(2);
+    step.philox = op.philox;
+  } else if (op.name == "aten::fill_") {
+    step.kind = ChainStep::Kind::kFill;
+    step.p0 = scalarArg(1);
+  } else if (op.name == "aten::zero_") {
+    step.kind = ChainStep::Kind::kZero;
+  } else if (op.name == "aten::detach" || op.name == "tdx::variable_data" ||
+             op.name == "aten::alias") {
+    step.kind = ChainStep::Kind::kPass;
+  } else {
+    TORCH_CHECK(false, "slice materialization: `", op.name,
+                "` is not a whole-tensor init op; materialize the tensor "
+                "fully instead.");
+  }
+  return step;
+}
+
+void applyShardStep(const ChainStep& step, at::Tensor& shard, int64_t start,
+                    int64_t end) {
+  switch (step.kind) {
+    case ChainStep::Kind::kFactory:
+    case ChainStep::Kind::kPass:
+      return;
+    case ChainStep::Kind::kZero:
+      shard.zero_();
+      return;
+    case ChainStep::Kind::kFill:
+      shard.fill_(step.p0);
+      return;
+    case ChainStep::Kind::kUniform:
+    case ChainStep::Kind::kNormal: {
+      TORCH_CHECK(step.philox.has_value(),
+                  "slice materialization: this RNG op carries no pinned "
+                  "Philox state (was it recorded by an older build?)");
+      static const auto uniform_shard =
+          c10::Dispatcher::singleton()
+              .findSchemaOrThrow("tdx::uniform_shard_", "")
+              .typed<at::Tensor&(at::Tensor&, int64_t, int64_t, double,
+                                 double, int64_t, int64_t)>();
+      static const auto normal_shard =
+          c10::Dispatcher::singleton()
+              .findSchemaOrThrow("tdx::normal_shard_", "")
+              .typed<at::Tensor&(at::Tensor&, int64_t, int64_t, double,
+                                 double, int64_t, int64_t)>();
+      const auto& [seed, offset] = *step.philox;
+      if (step.kind == ChainStep::Kind::kUniform) {
+        uniform_shard.call(shard, start, end, step.p0, step.p1,
+                           static_cast<int64_t>(seed),
+                           static_cast<int64_t>(offset));
+      } else {
+        normal_shard.call(shard, start, end, step.p0, step.p1,
+                          static_cast<int64_t>(seed),
+                          static_cast<int64_t>(offset));
+      }
+      return;
+    }
+  }
+}
+
+}  // namespace
+
+at::Tensor materializeTensorShard(const at::Tensor& tensor,
+                                  int64_t start_row,
+                                  int64_t end_row) {
+  auto* fake = asFake(tensor);
+  TORCH_CHECK_VALUE(fake != nullptr && getRecord(fake) != nullptr,
+                    "`tensor` is not a deferred tensor.");
+  const at::Tensor& meta = fake->meta_tensor();
+  TORCH_CHECK(meta.is_contiguous(),
+              "slice materialization requires a contiguous tensor");
+  const int64_t rows = meta.dim() == 0 ? 1 : meta.size(0);
+  TORCH_CHECK(0 <= start_row && start_row <= end_row && end_row <= rows,
+              "invalid row range [", start_row, ", ", end_row, ") for ",
+              rows, " rows");
+  const int64_t row_elems = rows == 0 ? 0 : meta.numel() / rows;
+
+  std::lock_guard<std::recursive_mutex> lock{tape_mutex};
+  auto rec = getRecord(fake);
+  auto nodes = buildCallStack(*rec);
+  std::vector<ChainStep> steps;
+  steps.reserve(nodes.size());
+  for (const auto& n : nodes) {
+    steps.push_back(classifyChainNode(*n));
+  }
+  TORCH_CHECK(!steps.empty() &&
+                  (steps.front().kind == ChainStep::Kind::kFactory ||
+                   steps.front().kind == ChainStep::Kind::kZero ||
+                   steps.front().kind == ChainStep::Kind::kFill),
+              "slice materialization: the tape does not start with a "
+              "factory op");
+
+  std::vector<int64_t> shard_sizes(meta.sizes().begin(), meta.sizes().end());
+  if (!shard_sizes.empty()) {
+    shard_sizes[0] = end_row - start_row;
+  }
+  at::Tensor shard = at::empty(
+      shard_sizes, at::TensorOptions()
+                       .dtype(meta.scalar_type())
+                       .device(fake->fake_device()));
+  const int64_t start = start_row * row_elems;
+  const int64_t end = end_row * row_elems;
+  for (const ChainStep& step : steps) {
+    applyShardStep(step, shard, start, end);
+  }
+  return shard;
 }
 
 at::Tensor materializeTensor(const at::Tensor& tensor) {

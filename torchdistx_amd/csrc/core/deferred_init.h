@@ -57,4 +57,19 @@ struct RecordInfo {
 // std::nullopt when `tensor` carries no deferred-init record.
 std::optional<RecordInfo> recordInfo(const at::Tensor& tensor);
 
+// Slice materialization: materializes rows [start_row, end_row) of the
+// deferred tensor's dim 0 WITHOUT materializing the rest, bitwise-equal to
+// the corresponding slice of a full materialization (per device type).
+// Requires the tensor's tape to be a "simple init chain" — a factory
+// (empty/zeros/ones/full) followed by whole-tensor in-place init ops
+// (uniform_/normal_/fill_/zero_) and aliasing pass-throughs
+// (detach/variable_data) — which is exactly what module constructors
+// record; throws a descriptive error otherwise so callers can fall back
+// to full materialization. This is the FSDP/TP init primitive: N ranks
+// materialize disjoint slices of a model larger than any single device,
+// with zero communication.
+at::Tensor materializeTensorShard(const at::Tensor& tensor,
+                                  int64_t start_row,
+                                  int64_t end_row);
+
 }  // namespace tdx

@@ -77,6 +77,9 @@ PYBIND11_MODULE(_C, m) {
   m.def("set_native_init", &tdx::setNativeInitEnabled,
         pybind11::arg("enabled"));
   m.def("native_init_enabled", &tdx::nativeInitEnabled);
+  m.def("set_native_init_cpu", &tdx::setNativeInitCpuEnabled,
+        pybind11::arg("enabled"));
+  m.def("native_init_cpu_enabled", &tdx::nativeInitCpuEnabled);
 
   m.def("record_info", [](const at::Tensor& t) -> pybind11::object {
     auto info = tdx::recordInfo(t);
@@ -100,4 +103,14 @@ PYBIND11_MODULE(_C, m) {
     }
     return out;
   });
+
+  m.def("materialize_tensor_shard",
+        [](const at::Tensor& t, int64_t start_row, int64_t end_row) {
+          at::Tensor out;
+          {
+            pybind11::gil_scoped_release release;
+            out = tdx::materializeTensorShard(t, start_row, end_row);
+          }
+          return out;
+        });
 }

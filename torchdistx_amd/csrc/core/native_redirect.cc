@@ -8,6 +8,7 @@ namespace tdx {
 namespace {
 
 std::atomic<bool> native_init_enabled{true};
+std::atomic<bool> native_init_cpu_enabled{false};
 
 struct Redirect {
   const char* aten_name;
@@ -39,6 +40,14 @@ bool nativeInitEnabled() noexcept {
   return native_init_enabled.load(std::memory_order_relaxed);
 }
 
+void setNativeInitCpuEnabled(bool enabled) noexcept {
+  native_init_cpu_enabled.store(enabled, std::memory_order_relaxed);
+}
+
+bool nativeInitCpuEnabled() noexcept {
+  return native_init_cpu_enabled.load(std::memory_order_relaxed);
+}
+
 bool tryNativeInitRedirect(
     const c10::OperatorHandle& op,
     torch::jit::Stack& stack,
@@ -64,7 +73,10 @@ bool tryNativeInitRedirect(
     return false;
   }
   const at::Tensor& self = stack.front().toTensor();
-  if (!self.defined() || !self.is_cuda() || !self.is_contiguous()) {
+  if (!self.defined() || !self.is_contiguous()) {
+    return false;
+  }
+  if (!self.is_cuda() && !(self.is_cpu() && nativeInitCpuEnabled())) {
     return false;
   }
   if (redirect->needs_float_dtype) {

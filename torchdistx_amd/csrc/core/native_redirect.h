@@ -35,4 +35,11 @@ bool tryNativeInitRedirect(
 void setNativeInitEnabled(bool enabled) noexcept;
 bool nativeInitEnabled() noexcept;
 
+// Opt-in: also redirect CPU-target init ops to the tdx CPU reference
+// implementations (pinned-Philox layout). Off by default so plain CPU
+// replay stays bitwise-equal to eager construction; turn on when CPU
+// materialization must be slice-consistent with materialize_tensor_shard.
+void setNativeInitCpuEnabled(bool enabled) noexcept;
+bool nativeInitCpuEnabled() noexcept;
+
 }  // namespace tdx

@@ -142,12 +142,55 @@ struct VecTraits<__half> {
 
 enum class Dist { kUniform, kNormal };
 
-// One kernel for both distributions: each iteration runs ONE philox10 with
-// counter = the 16-byte group index and turns its 128 bits into one
-// 16-byte store — 4 fp32 samples (24-bit uniforms) or 8 bf16/fp16 samples
-// (16-bit uniforms, matched to the output mantissa). IdxT is uint32_t
-// whenever the group count fits (the hot case; 64-bit address math costs
-// ~2x VALU on this loop).
+// Computes the VecTraits<T>::kElems values of element group `g` — ONE
+// philox10 with counter = the 16-byte group index turned into 4 fp32
+// samples (24-bit uniforms) or 8 bf16/fp16 samples (16-bit uniforms,
+// matched to the output mantissa). Shared by the full-tensor and the
+// shard kernels so a shard is bitwise a slice of the full tensor.
+template <typename T, Dist kDist>
+__device__ __forceinline__ void rngGroupValues(uint64_t g,
+                                               float a,
+                                               float b,
+                                               uint64_t seed,
+                                               uint64_t offset,
+                                               float* vals) {
+  uint4 bits = philox10(seed, g, offset);
+  if constexpr (VecTraits<T>::kElems == 4) {
+    float u[4] = {u32_to_uniform(bits.x), u32_to_uniform(bits.y),
+                  u32_to_uniform(bits.z), u32_to_uniform(bits.w)};
+    if constexpr (kDist == Dist::kUniform) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        vals[j] = fmaf(u[j], b, a);
+      }
+    } else {
+      float2 n01 = box_muller(u[0], u[1]);
+      float2 n23 = box_muller(u[2], u[3]);
+      vals[0] = fmaf(n01.x, b, a);
+      vals[1] = fmaf(n01.y, b, a);
+      vals[2] = fmaf(n23.x, b, a);
+      vals[3] = fmaf(n23.y, b, a);
+    }
+  } else {
+    uint32_t words[4] = {bits.x, bits.y, bits.z, bits.w};
+    if constexpr (kDist == Dist::kUniform) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        vals[j * 2 + 0] = fmaf(u16_to_uniform(words[j]), b, a);
+        vals[j * 2 + 1] = fmaf(u16_to_uniform(words[j] >> 16), b, a);
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float2 nj = box_muller(u16_to_uniform(words[j]),
+                               u16_to_uniform(words[j] >> 16));
+        vals[j * 2 + 0] = fmaf(nj.x, b, a);
+        vals[j * 2 + 1] = fmaf(nj.y, b, a);
+      }
+    }
+  }
+}
+
 template <typename T, Dist kDist, typename IdxT>
 __global__ void rng_kernel(T* __restrict__ out,
                            IdxT n,
@@ -163,43 +206,9 @@ __global__ void rng_kernel(T* __restrict__ out,
 
   for (IdxT g = blockIdx.x * static_cast<IdxT>(blockDim.x) + threadIdx.x;
        g < n_groups; g += stride) {
-    uint4 bits = philox10(seed, static_cast<uint64_t>(g), offset);
     float vals[kElems];
-    if constexpr (kElems == 4) {
-      float u[4] = {u32_to_uniform(bits.x), u32_to_uniform(bits.y),
-                    u32_to_uniform(bits.z), u32_to_uniform(bits.w)};
-      if constexpr (kDist == Dist::kUniform) {
-#pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          vals[j] = fmaf(u[j], b, a);
-        }
-      } else {
-        float2 n01 = box_muller(u[0], u[1]);
-        float2 n23 = box_muller(u[2], u[3]);
-        vals[0] = fmaf(n01.x, b, a);
-        vals[1] = fmaf(n01.y, b, a);
-        vals[2] = fmaf(n23.x, b, a);
-        vals[3] = fmaf(n23.y, b, a);
-      }
-    } else {
-      uint32_t words[4] = {bits.x, bits.y, bits.z, bits.w};
-      if constexpr (kDist == Dist::kUniform) {
-#pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          vals[j * 2 + 0] = fmaf(u16_to_uniform(words[j]), b, a);
-          vals[j * 2 + 1] = fmaf(u16_to_uniform(words[j] >> 16), b, a);
-        }
-      } else {
-#pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          float2 nj = box_muller(u16_to_uniform(words[j]),
-                                 u16_to_uniform(words[j] >> 16));
-          vals[j * 2 + 0] = fmaf(nj.x, b, a);
-          vals[j * 2 + 1] = fmaf(nj.y, b, a);
-        }
-      }
-    }
-
+    rngGroupValues<T, kDist>(static_cast<uint64_t>(g), a, b, seed, offset,
+                             vals);
     const IdxT base = g * kElems;
     if (base + kElems <= n) {
       Vec v;
@@ -351,6 +360,112 @@ at::Tensor& tdx_normal_(at::Tensor& self,
   TORCH_CHECK(std >= 0.0, "normal_ expects std >= 0.0, but found std=", std);
   launchRng<Dist::kNormal>(self, mean, std, generator, seed, offset);
   return self;
+}
+
+// Shard kernel: writes elements [start, end) of the virtual full tensor
+// into out[0 .. end-start). Interior groups use full 16-byte stores;
+// boundary groups store elementwise.
+template <typename T, Dist kDist>
+__global__ void rng_shard_kernel(T* __restrict__ out,
+                                 int64_t start,
+                                 int64_t end,
+                                 float a,
+                                 float b,
+                                 uint64_t seed,
+                                 uint64_t offset) {
+  constexpr int kElems = VecTraits<T>::kElems;
+  using Vec = typename VecTraits<T>::Vec;
+
+  const int64_t g_first = start / kElems;
+  const int64_t g_last = (end + kElems - 1) / kElems;
+  const int64_t stride = static_cast<int64_t>(gridDim.x) * blockDim.x;
+
+  for (int64_t g = g_first +
+           blockIdx.x * static_cast<int64_t>(blockDim.x) + threadIdx.x;
+       g < g_last; g += stride) {
+    float vals[kElems];
+    rngGroupValues<T, kDist>(static_cast<uint64_t>(g), a, b, seed, offset,
+                             vals);
+    const int64_t base = g * kElems;
+    if (base >= start && base + kElems <= end) {
+      Vec v;
+      T* vp = reinterpret_cast<T*>(&v);
+#pragma unroll
+      for (int j = 0; j < kElems; ++j) {
+        vp[j] = from_float<T>(vals[j]);
+      }
+      *reinterpret_cast<Vec*>(out + (base - start)) = v;
+    } else {
+      const int64_t lo = base > start ? base : start;
+      const int64_t hi = base + kElems < end ? base + kElems : end;
+      for (int64_t e = lo; e < hi; ++e) {
+        out[e - start] = from_float<T>(vals[e - base]);
+      }
+    }
+  }
+}
+
+template <Dist kDist>
+void launchRngShard(at::Tensor& shard,
+                    int64_t start,
+                    int64_t end,
+                    double p0,
+                    double p1,
+                    int64_t seed,
+                    int64_t offset) {
+  TORCH_CHECK(shard.is_contiguous(),
+              "tdx shard kernels require contiguous tensors");
+  TORCH_CHECK(shard.numel() == end - start,
+              "shard numel must equal end - start");
+  if (shard.numel() == 0) {
+    return;
+  }
+  float a = static_cast<float>(p0);
+  float b = kDist == Dist::kUniform ? static_cast<float>(p1 - p0)
+                                    : static_cast<float>(p1);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  auto launch = [&](auto type_tag) {
+    using T = decltype(type_tag);
+    const int64_t n_groups =
+        (end + VecTraits<T>::kElems - 1) / VecTraits<T>::kElems -
+        start / VecTraits<T>::kElems;
+    hipLaunchKernelGGL((rng_shard_kernel<T, kDist>),
+                       dim3(numBlocks(n_groups)), dim3(kBlock), 0,
+                       stream.stream(),
+                       reinterpret_cast<T*>(shard.data_ptr()), start, end, a,
+                       b, static_cast<uint64_t>(seed),
+                       static_cast<uint64_t>(offset));
+    C10_HIP_KERNEL_LAUNCH_CHECK();
+  };
+  switch (shard.scalar_type()) {
+    case at::kFloat:
+      launch(float{});
+      break;
+    case at::kBFloat16:
+      launch(__hip_bfloat16{});
+      break;
+    case at::kHalf:
+      launch(__half{});
+      break;
+    default:
+      TORCH_CHECK(false, "tdx shard kernels support float32/bf16/fp16, got ",
+                  shard.scalar_type());
+  }
+}
+
+at::Tensor& tdx_uniform_shard_(at::Tensor& shard, int64_t start, int64_t end,
+                               double from, double to, int64_t seed,
+                               int64_t offset) {
+  launchRngShard<Dist::kUniform>(shard, start, end, from, to, seed, offset);
+  return shard;
+}
+
+at::Tensor& tdx_normal_shard_(at::Tensor& shard, int64_t start, int64_t end,
+                              double mean, double std, int64_t seed,
+                              int64_t offset) {
+  TORCH_CHECK(std >= 0.0, "normal_ expects std >= 0.0, but found std=", std);
+  launchRngShard<Dist::kNormal>(shard, start, end, mean, std, seed, offset);
+  return shard;
 }
 
 at::Tensor& tdx_fill_(at::Tensor& self, const at::Scalar& value) {
@@ -515,28 +630,16 @@ at::Tensor& tdx_zero_(at::Tensor& self) {
   return self;
 }
 
-TORCH_LIBRARY(tdx, m) {
-  m.def(
-      "uniform_(Tensor(a!) self, float from=0., float to=1., *, "
-      "Generator? generator=None, int? seed=None, int? offset=None) "
-      "-> Tensor(a!)");
-  m.def(
-      "normal_(Tensor(a!) self, float mean=0., float std=1., *, "
-      "Generator? generator=None, int? seed=None, int? offset=None) "
-      "-> Tensor(a!)");
-  m.def("fill_(Tensor(a!) self, Scalar value) -> Tensor(a!)");
-  m.def("zero_(Tensor(a!) self) -> Tensor(a!)");
-  m.def(
-      "copy_(Tensor(a!) self, Tensor src, bool non_blocking=False) "
-      "-> Tensor(a!)");
-}
-
+// Schemas are defined by the core extension (csrc/core/tdx_ops.cc); this
+// extension contributes the CUDA implementations.
 TORCH_LIBRARY_IMPL(tdx, CUDA, m) {
   m.impl("uniform_", tdx_uniform_);
   m.impl("normal_", tdx_normal_);
   m.impl("fill_", tdx_fill_);
   m.impl("zero_", tdx_zero_);
   m.impl("copy_", tdx_copy_);
+  m.impl("uniform_shard_", tdx_uniform_shard_);
+  m.impl("normal_shard_", tdx_normal_shard_);
 }
 
 }  // namespace

@@ -171,3 +171,63 @@ def materialize_experts_sharded(
 
     materialize_module(module, check_fn=lambda sub: id(sub) not in skip_ids)
     return owners
+
+
+def materialize_tensor_shard(
+    tensor: torch.Tensor, start_row: int, end_row: int
+) -> torch.Tensor:
+    """Materializes rows [start_row, end_row) of a deferred tensor's dim 0
+    without touching the rest — bitwise-equal to the corresponding slice of
+    a full native materialization, at shard cost. See
+    csrc/core/deferred_init.h (materializeTensorShard) for the supported
+    tape shapes; unsupported tapes raise, so callers can fall back to
+    `materialize_tensor` + slicing."""
+    try:
+        shard = _C.materialize_tensor_shard(tensor, start_row, end_row)
+    except RuntimeError as e:
+        if "slice materialization" not in str(e):
+            raise
+        # Not a simple init chain (e.g. a computed buffer like a RoPE
+        # cache): fall back to full materialization and slice. Costs the
+        # full tensor once; fine for the small computed buffers this
+        # covers, and cross-rank consistency still holds (every rank
+        # replays the same tape).
+        full = _C.materialize_tensor(tensor)
+        shard = full[start_row:end_row].clone() if full.dim() > 0 else full
+    if tensor.requires_grad:
+        shard.requires_grad_(True)
+    return shard
+
+
+def materialize_module_dim0_sharded(
+    module: Module,
+    rank: Optional[int] = None,
+    world_size: Optional[int] = None,
+) -> Dict[str, torch.Tensor]:
+    """FSDP/TP-style init: every parameter and buffer of the deferred
+    ``module`` is split contiguously along dim 0 across ``world_size``
+    ranks and only this rank's slice is materialized. Zero communication;
+    slices are bitwise-consistent with a full materialization through the
+    native kernels, so concatenating all ranks' results reconstructs the
+    exact full model. N ranks can therefore initialize a model larger than
+    any single device (e.g. Llama-3-405B, 812 GB bf16, across 8 GPUs).
+
+    Returns {fully-qualified tensor name -> local shard}. The module's own
+    entries are left fake (they describe the full tensors).
+    """
+    if rank is None:
+        rank = dist.get_rank() if dist.is_initialized() else 0
+    if world_size is None:
+        world_size = dist.get_world_size() if dist.is_initialized() else 1
+
+    shards: Dict[str, torch.Tensor] = {}
+    for name, t in list(module.named_parameters()) + list(
+        module.named_buffers()
+    ):
+        if not _C.can_materialize(t):
+            continue
+        rows = t.shape[0] if t.dim() > 0 else 1
+        start = rank * rows // world_size
+        end = (rank + 1) * rows // world_size
+        shards[name] = materialize_tensor_shard(t, start, end)
+    return shards
