@@ -38,7 +38,7 @@ def get_args():
     p.add_argument("--model", type=str, default=None,
                    help="config name (default: llama3-70b on GPU, tiny on CPU)")
     p.add_argument("--mode", type=str, default="replicate",
-                   choices=["replicate", "shard", "broadcast"])
+                   choices=["replicate", "shard", "broadcast", "slice"])
     p.add_argument("--dtype", type=str, default="bf16",
                    choices=["bf16", "fp32", "fp16"])
     return p.parse_args()
@@ -74,7 +74,10 @@ def main():
 
     from torchdistx_amd import deferred_init, is_deferred, _kernels
     from torchdistx_amd.models import CONFIGS, build_model
-    from torchdistx_amd.parallel import materialize_module_distributed
+    from torchdistx_amd.parallel import (
+        materialize_module_dim0_sharded,
+        materialize_module_distributed,
+    )
 
     if use_cuda and not _kernels.available():
         print("FATAL: torchdistx_amd._K (CDNA4 kernels) not loaded on a GPU box",
@@ -89,9 +92,16 @@ def main():
     def one_step(seed: int) -> None:
         torch.manual_seed(seed)
         module = deferred_init(build_model, cfg, device=device, dtype=dtype)
-        materialize_module_distributed(module, mode=args.mode)
-        if args.mode != "shard":
-            assert not is_deferred(module), "materialization incomplete"
+        if args.mode == "slice":
+            # sub-tensor FSDP/TP-style init: this rank materializes its
+            # contiguous dim-0 row-slice of every parameter.
+            shards = materialize_module_dim0_sharded(module, rank, world)
+            assert shards
+            del shards
+        else:
+            materialize_module_distributed(module, mode=args.mode)
+            if args.mode != "shard":
+                assert not is_deferred(module), "materialization incomplete"
         # Reference-counting frees the module and its tape immediately (the
         # tape's ownership graph is acyclic by design); no gc.collect().
         del module
