@@ -52,12 +52,8 @@ __device__ __forceinline__ void storef(void* p, DT dt, int64_t i, float v) {
   }
 }
 
-// Vector traits: 8 elements per thread per iteration, 16-byte stores for
+// Vector packs: 8 elements per thread per iteration, 16-byte accesses for
 // 16-bit types, 2x16-byte for fp32.
-template <typename T>
-struct Pack8 {
-  T v[8];
-};
 template <typename T>
 struct alignas(16) Pack8Aligned {
   T v[8];
