@@ -38,7 +38,7 @@ def get_args():
     p.add_argument("--model", type=str, default=None,
                    help="config name (default: llama3-70b on GPU, tiny on CPU)")
     p.add_argument("--mode", type=str, default="replicate",
-                   choices=["replicate", "shard", "broadcast", "slice"])
+                   choices=["replicate", "shard", "broadcast", "allgather", "slice"])
     p.add_argument("--dtype", type=str, default="bf16",
                    choices=["bf16", "fp32", "fp16"])
     return p.parse_args()

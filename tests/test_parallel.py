@@ -227,3 +227,31 @@ def test_slice_materialization_complex_tape_fallback() -> None:
     assert torch.equal(shard.detach(), expected)
     tail = materialize_tensor_shard(m.p, 4, 8)
     assert torch.equal(tail.detach(), torch.zeros(4, 8))
+
+
+def _allgather_worker(rank, world):
+    from torchdistx_amd import _C, deferred_init
+    from torchdistx_amd.deferred_init import materialize_module
+    from torchdistx_amd.models import TINY, build_model
+    from torchdistx_amd.parallel import materialize_module_distributed
+
+    _C.set_native_init_cpu(True)
+    try:
+        torch.manual_seed(9)
+        m = deferred_init(build_model, TINY, device="cpu", dtype=torch.float32)
+        materialize_module_distributed(m, mode="allgather")
+        torch.manual_seed(9)
+        ref = deferred_init(build_model, TINY, device="cpu", dtype=torch.float32)
+        materialize_module(ref)
+        return all(
+            torch.equal(a.detach(), b.detach())
+            for (_, a), (_, b) in zip(m.named_parameters(), ref.named_parameters())
+        )
+    finally:
+        _C.set_native_init_cpu(False)
+
+
+def test_allgather_mode_reconstructs_native_full() -> None:
+    # Per-rank slice init + all-gather equals a full native materialization
+    # bitwise on every rank (uneven splits padded internally).
+    assert all(run_distributed(_allgather_worker, 2))

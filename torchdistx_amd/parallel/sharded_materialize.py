@@ -81,8 +81,13 @@ def materialize_module_distributed(
 
     See the module docstring for the mode semantics.
     """
-    if mode not in ("replicate", "shard", "broadcast"):
+    if mode not in ("replicate", "shard", "broadcast", "allgather"):
         raise ValueError(f"unknown materialization mode: {mode!r}")
+
+    if mode == "allgather" and dist.is_initialized():
+        return _materialize_allgather(
+            module, process_group, buffers_only, check_fn
+        )
 
     if mode == "replicate" or not dist.is_initialized():
         from torchdistx_amd.deferred_init import materialize_module
@@ -171,6 +176,60 @@ def materialize_experts_sharded(
 
     materialize_module(module, check_fn=lambda sub: id(sub) not in skip_ids)
     return owners
+
+
+def _materialize_allgather(
+    module: Module,
+    process_group: Optional[dist.ProcessGroup],
+    buffers_only: bool,
+    check_fn: Optional[Callable[[Module], bool]],
+) -> Dict[str, int]:
+    """Every rank slice-materializes 1/world of each tensor's rows, then an
+    async all-gather reconstructs full bitwise-identical replicas on every
+    rank. Exists for completeness and comparison: on xGMI, `replicate`
+    (local regeneration at HBM write speed) beats moving weights over the
+    ~153 GB/s links — measure both and see. Init work per rank is 1/world;
+    the collectives overlap with the remaining slice kernels."""
+    group = process_group or dist.group.WORLD
+    world = dist.get_world_size(group)
+    rank = dist.get_rank(group)
+
+    entries = _named_deferred_tensors(module, buffers_only)
+    if check_fn is not None:
+        entries = [e for e in entries if check_fn(e[0])]
+
+    handles = []
+    for submodule, key, tensor, is_param in entries:
+        rows = tensor.shape[0] if tensor.dim() > 0 else 1
+        # Equal-size slots so all_gather_into_tensor applies; the last
+        # slots may be partly past the end and are narrowed away below.
+        slot = -(-rows // world)  # ceil
+        start = min(rank * slot, rows)
+        end = min(start + slot, rows)
+        shard = materialize_tensor_shard(tensor, start, end)
+        slot_shape = (slot,) + tuple(tensor.shape[1:])
+        padded = torch.zeros(slot_shape, dtype=tensor.dtype,
+                             device=tensor.device)
+        if end > start:
+            padded[: end - start] = shard.detach()
+        full_padded = torch.empty((slot * world,) + tuple(tensor.shape[1:]),
+                                  dtype=tensor.dtype, device=tensor.device)
+        handles.append(
+            dist.all_gather_into_tensor(
+                full_padded, padded, group=group, async_op=True
+            )
+        )
+        mat = full_padded.narrow(0, 0, rows).view(tensor.shape)
+        if is_param:
+            mat.requires_grad_(tensor.requires_grad)
+            mat = _restore_class(tensor, mat)
+            submodule._parameters[key] = mat
+        else:
+            submodule._buffers[key] = mat
+
+    for h in handles:
+        h.wait()
+    return {}
 
 
 def materialize_tensor_shard(
