@@ -1,0 +1,34 @@
+# Tape observability (torchdistx_amd.utils.tape).
+
+import torch
+from torch.nn import Linear
+
+from torchdistx_amd import deferred_init, materialize_module
+from torchdistx_amd.models import TINY, build_model
+from torchdistx_amd.utils import describe_module, record_info
+
+
+def test_record_info_lifecycle() -> None:
+    m = deferred_init(Linear, 4, 4)
+    info = record_info(m.weight)
+    assert info is not None
+    assert not info["materialized"]
+    assert info["pending_ops"] >= 2  # empty + uniform_ at minimum
+    assert info["op_name"]
+
+    materialize_module(m)
+    assert record_info(m.weight) is None  # real tensors carry no record
+
+
+def test_describe_module_counts() -> None:
+    m = deferred_init(build_model, TINY)
+    d = describe_module(m)
+    n_tensors = sum(1 for _ in m.parameters()) + sum(1 for _ in m.buffers())
+    assert d["n_recorded_tensors"] == n_tensors
+    assert d["n_awaiting_materialization"] == n_tensors
+    assert d["total_pending_ops"] >= n_tensors
+
+    materialize_module(m)
+    d2 = describe_module(m)
+    assert d2["n_awaiting_materialization"] == 0
+    assert d2["n_recorded_tensors"] == 0  # records dropped with the fakes
