@@ -59,20 +59,27 @@ PYBIND11_MODULE(_C, m) {
     // translators (pybind runs translators in reverse registration order).
   });
 
-  m.def("enter_fake_mode", &enterFakeModeBinding, pybind11::arg("fake_cuda"));
-  m.def("leave_fake_mode", &leaveFakeModeBinding);
+  m.def("enter_fake_mode", &enterFakeModeBinding, pybind11::arg("fake_cuda"),
+        "Enter fake-tensor mode on this thread (nestable). With fake_cuda, "
+        "fake \"cuda\" tensors can be built without a GPU runtime.");
+  m.def("leave_fake_mode", &leaveFakeModeBinding,
+        "Leave the innermost fake mode.");
 
-  m.def("is_fake",
-        [](const at::Tensor& t) { return tdx::isFake(t); });
+  m.def("is_fake", [](const at::Tensor& t) { return tdx::isFake(t); },
+        "Whether the tensor is a storage-less fake tensor.");
 
-  m.def("meta_like",
-        [](const at::Tensor& t) { return tdx::metaLike(t); });
+  m.def("meta_like", [](const at::Tensor& t) { return tdx::metaLike(t); },
+        "Detached meta tensor with the fake tensor's metadata; raises "
+        "ValueError for non-fake inputs.");
 
-  m.def("enter_deferred_init", &tdx::enterDeferredInit);
-  m.def("leave_deferred_init", &tdx::leaveDeferredInit);
+  m.def("enter_deferred_init", &tdx::enterDeferredInit,
+        "Enter deferred-init recording on this thread (nestable).");
+  m.def("leave_deferred_init", &tdx::leaveDeferredInit,
+        "Leave the innermost deferred-init context.");
 
   m.def("can_materialize",
-        [](const at::Tensor& t) { return tdx::canMaterialize(t); });
+        [](const at::Tensor& t) { return tdx::canMaterialize(t); },
+        "Whether the tensor is fake AND carries a replayable tape record.");
 
   m.def("set_native_init", &tdx::setNativeInitEnabled,
         pybind11::arg("enabled"));
@@ -102,7 +109,8 @@ PYBIND11_MODULE(_C, m) {
       out = tdx::materializeTensor(t);
     }
     return out;
-  });
+  }, "Replays the tensor's tape segment (GIL released); identity for real "
+     "tensors, stable object across repeated calls and aliases.");
 
   m.def("materialize_tensor_shard",
         [](const at::Tensor& t, int64_t start_row, int64_t end_row) {
@@ -112,5 +120,8 @@ PYBIND11_MODULE(_C, m) {
             out = tdx::materializeTensorShard(t, start_row, end_row);
           }
           return out;
-        });
+        },
+        "Materializes rows [start_row, end_row) of the deferred tensor's "
+        "dim 0 alone, bitwise-equal to that slice of a full native "
+        "materialization (see docs/distributed_materialization.md).");
 }
