@@ -90,8 +90,8 @@ def _run_program(ops, device="cpu"):
                 fam.append(fam[op[1]] if base.is_contiguous() else len(fam))
         elif kind == "select":
             base = ts[op[1]]
-            if base.dim() == 0:
-                ts.append(base.reshape(1))
+            if base.dim() == 0 or base.shape[0] == 0:
+                ts.append(base.reshape(base.numel()))
             else:
                 ts.append(base[min(op[2], base.shape[0] - 1)])
             fam.append(fam[op[1]])

@@ -168,12 +168,20 @@ struct InputSlot {
   bool inference = false;
 };
 
+struct AliasGroup;
+
 struct OpNode {
   uint64_t op_nr = 0;
   std::optional<RecordedOp> op;          // dropped after replay
   std::vector<InputSlot> input_slots;    // tensor visit order over args
   std::vector<c10::Storage> output_storages;  // meta storages, visit order
   std::vector<std::weak_ptr<OpNode>> dependents;
+  // The storage families this node writes (weak: groups own nodes, never
+  // the reverse). Replay collection closes over each collected writer's
+  // chronological prefix in these families, so a writer forced into a
+  // replay set (e.g. by the clobber rule) can never run before an earlier
+  // writer of the same storage.
+  std::vector<std::weak_ptr<AliasGroup>> write_groups;
   std::vector<at::Tensor> outputs;  // real tensors after replay
   bool materialized = false;
 };
@@ -311,6 +319,7 @@ void recordOp(std::string name,
     if (rec->group->writers.empty() ||
         rec->group->writers.back() != node) {
       rec->group->writers.push_back(node);
+      node->write_groups.emplace_back(rec->group);
     }
   });
 
@@ -648,6 +657,17 @@ struct CallStack {
       }
       for (const auto& h : slot.history) {
         addWithDeps(h);
+      }
+    }
+    // Chronological-prefix closure: every earlier writer of every family
+    // this node writes must replay before it.
+    for (const auto& wg : n->write_groups) {
+      if (auto group = wg.lock()) {
+        for (const auto& w : group->writers) {
+          if (w->op_nr < n->op_nr) {
+            addWithDeps(w);
+          }
+        }
       }
     }
   }
