@@ -117,3 +117,15 @@ def test_fake_mode_nesting() -> None:
         b = torch.ones([2])
     assert is_fake(a) and is_fake(b)
     assert not is_fake(torch.ones([2]))
+
+
+def test_backward_through_fake_tensors() -> None:
+    # Autograd runs above the Fake key, so backward works end-to-end on
+    # fake tensors and produces fake gradients.
+    with fake_mode():
+        x = torch.randn(4, 4, requires_grad=True)
+        y = (x * 2).sum()
+        y.backward()
+    assert x.grad is not None
+    assert is_fake(x.grad)
+    assert x.grad.shape == (4, 4)
