@@ -6,7 +6,7 @@ import torch
 
 
 def test_alias_submodules_and_api() -> None:
-    import torchdistx
+    import torchdistx  # noqa: F401  (import is the point)
     from torchdistx import deferred_init, fake
 
     m = deferred_init.deferred_init(torch.nn.Linear, 4, 4)

@@ -3,7 +3,6 @@
 
 import pytest
 import torch
-import torch.distributed as dist
 
 from tests._dist_utils import run_distributed
 from torchdistx_amd.parallel import assign_owners

@@ -1,6 +1,5 @@
 # Tape observability (torchdistx_amd.utils.tape).
 
-import torch
 from torch.nn import Linear
 
 from torchdistx_amd import deferred_init, materialize_module
