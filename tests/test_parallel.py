@@ -254,3 +254,37 @@ def test_allgather_mode_reconstructs_native_full() -> None:
     # Per-rank slice init + all-gather equals a full native materialization
     # bitwise on every rank (uneven splits padded internally).
     assert all(run_distributed(_allgather_worker, 2))
+
+
+def _dtensor_worker(rank, world):
+    from torch.distributed.device_mesh import init_device_mesh
+
+    from torchdistx_amd import _C, deferred_init
+    from torchdistx_amd.deferred_init import materialize_module
+    from torchdistx_amd.models import TINY, build_model
+    from torchdistx_amd.parallel import materialize_module_dtensor
+
+    _C.set_native_init_cpu(True)
+    try:
+        mesh = init_device_mesh("cpu", (world,))
+        torch.manual_seed(2)
+        m = deferred_init(build_model, TINY, device="cpu", dtype=torch.float32)
+        dts = materialize_module_dtensor(m, mesh)
+
+        torch.manual_seed(2)
+        ref = deferred_init(build_model, TINY, device="cpu", dtype=torch.float32)
+        materialize_module(ref)
+        refp = dict(list(ref.named_parameters()) + list(ref.named_buffers()))
+        return all(
+            torch.equal(dt.full_tensor(), refp[name].detach())
+            for name, dt in dts.items()
+        )
+    finally:
+        _C.set_native_init_cpu(False)
+
+
+def test_dtensor_materialization_matches_full() -> None:
+    # FSDP2-style: each rank materializes only its Shard(0) chunk straight
+    # into DTensors; the assembled full tensors equal a native full
+    # materialization bitwise.
+    assert all(run_distributed(_dtensor_worker, 2))

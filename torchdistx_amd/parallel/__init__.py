@@ -2,6 +2,7 @@ from torchdistx_amd.parallel.sharded_materialize import (  # noqa: F401
     assign_owners,
     materialize_experts_sharded,
     materialize_module_dim0_sharded,
+    materialize_module_dtensor,
     materialize_module_distributed,
     materialize_tensor_shard,
 )

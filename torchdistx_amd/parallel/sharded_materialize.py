@@ -290,3 +290,37 @@ def materialize_module_dim0_sharded(
         end = (rank + 1) * rows // world_size
         shards[name] = materialize_tensor_shard(t, start, end)
     return shards
+
+
+def materialize_module_dtensor(
+    module: Module,
+    device_mesh,
+) -> Dict[str, "torch.Tensor"]:
+    """FSDP2-era init: every parameter/buffer of the deferred ``module``
+    materializes directly as a ``DTensor`` sharded on dim 0 over the given
+    1-D device mesh — each rank slice-materializes only its local chunk
+    (torch.chunk split semantics, matching ``distribute_tensor``), so the
+    full model never exists on any device. Returns {fqn -> DTensor}."""
+    from torch.distributed.tensor import DTensor, Shard
+
+    if device_mesh.ndim != 1:
+        raise ValueError("materialize_module_dtensor expects a 1-D mesh")
+    world = device_mesh.size()
+    rank = device_mesh.get_local_rank()
+
+    out: Dict[str, torch.Tensor] = {}
+    for name, t in list(module.named_parameters()) + list(
+        module.named_buffers()
+    ):
+        if not _C.can_materialize(t):
+            continue
+        rows = t.shape[0] if t.dim() > 0 else 1
+        slot = -(-rows // world)  # torch.chunk: ceil-size slots, short tail
+        start = min(rank * slot, rows)
+        end = min(start + slot, rows)
+        local = materialize_tensor_shard(t, start, end)
+        out[name] = DTensor.from_local(
+            local, device_mesh, [Shard(0)], run_check=False,
+            shape=t.shape, stride=t.stride(),
+        )
+    return out
