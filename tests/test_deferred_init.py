@@ -311,3 +311,19 @@ def test_beyond_hardware_scale_instantiation() -> None:
     assert n == LLAMA3_405B.n_params
     assert n > 400e9
     assert m.blocks[0].attn.wq.weight.shape == (16384, 16384)
+
+
+def test_external_tensor_copy_into_param() -> None:
+    # The checkpoint-loading flow: copying an external (real) tensor into a
+    # deferred parameter records and replays faithfully.
+    ext = torch.arange(12.0).reshape(3, 4)
+
+    class M(Module):
+        def __init__(self):
+            super().__init__()
+            self.p = Parameter(torch.empty(3, 4))
+            self.p.data.copy_(ext)
+
+    m = deferred_init(M)
+    p = materialize_tensor(cast(Tensor, m.p))
+    assert torch.equal(p.detach(), ext)
