@@ -25,16 +25,22 @@ def describe_module(module: Module) -> Dict:
     tensors = {}
     total_pending = 0
     n_deferred = 0
+    pending_bytes = 0
     for name, t in list(module.named_parameters()) + list(module.named_buffers()):
         info = _C.record_info(t)
         if info is None:
             continue
+        info = dict(info)
+        info["nbytes"] = t.numel() * t.element_size()
         tensors[name] = info
-        n_deferred += 0 if info["materialized"] else 1
+        if not info["materialized"]:
+            n_deferred += 1
+            pending_bytes += info["nbytes"]
         total_pending += info["pending_ops"]
     return {
         "n_recorded_tensors": len(tensors),
         "n_awaiting_materialization": n_deferred,
         "total_pending_ops": total_pending,
+        "pending_bytes": pending_bytes,
         "tensors": tensors,
     }
