@@ -7,11 +7,14 @@
 //   VariableHooks proxy                  — :889-948, :1050-1128
 //
 // The tape is re-designed rather than translated: one mutable TensorRecord
-// per fake impl (restamped in place so view keep-alives follow in-place
-// rewrites), visit-order input slots instead of nulled stack entries, and a
-// single fixpoint set-collection covering the reference's three collection
-// rules (dependency closure, same-storage writers up to the last aliasing
-// writer, and readers whose inputs a later replayed writer would clobber).
+// per fake impl (restamped in place), per-storage AliasGroups owning the
+// family's writer nodes (an acyclic replacement for the reference's view
+// keep-alives), record-time history snapshots for cross-family reads, and
+// a replay collection that takes a family's full writer set, closes over
+// dependencies + each writer's chronological prefix, and pulls in (to
+// fixpoint) readers whose inputs a collected later writer would clobber.
+// Validated differentially against eager execution by tests/test_tape_fuzz
+// (thousands of random aliasing programs).
 
 #include "deferred_init.h"
 
@@ -24,7 +27,6 @@
 #include <unordered_set>
 #include <vector>
 
-#include <ATen/CPUGeneratorImpl.h>
 #include <ATen/Context.h>
 #include <ATen/ThreadLocalState.h>
 #include <ATen/core/dispatch/Dispatcher.h>
