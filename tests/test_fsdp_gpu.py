@@ -85,3 +85,19 @@ def test_slowmo_hook_registers_on_fsdp(nccl_world1) -> None:
     torch.cuda.synchronize()
     sd = optim.state_dict()
     assert sd["step"] == 3
+
+
+def test_get_num_modules_counts_nested_fsdp(nccl_world1) -> None:
+    # Reference parity: get_num_modules counts nested FSDP units including
+    # the root (reference gossip_grad.py:319-331, tested at
+    # test_comm_hooks_fsdp.py:641-651).
+    from torch.distributed.fsdp import FullyShardedDataParallel as FSDP
+
+    from torchdistx_amd.gossip_grad import get_num_modules
+
+    inner = FSDP(torch.nn.Linear(4, 4).cuda())
+    outer = FSDP(
+        torch.nn.Sequential(inner, torch.nn.Linear(4, 4).cuda())
+    )
+    assert get_num_modules(outer) == 2
+    assert get_num_modules(inner) == 1
