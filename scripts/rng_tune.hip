@@ -247,6 +247,66 @@ __global__ void normal_bf16_log2(__hip_bfloat16* __restrict__ out,
   }
 }
 
+// Split-phase x2: both philox results first, then all transcendental
+// chains back-to-back so the scheduler can interleave them.
+__global__ void normal_bf16_split2(__hip_bfloat16* __restrict__ out,
+                                   uint32_t n8, float a, float b,
+                                   uint64_t seed, uint64_t offset) {
+  const uint32_t stride = gridDim.x * blockDim.x;
+  const uint32_t half = (n8 + 1) / 2;
+  for (uint32_t t = blockIdx.x * blockDim.x + threadIdx.x; t < half;
+       t += stride) {
+    uint32_t g0 = t, g1 = t + half;
+    uint4 b0 = philox10(seed, g0, offset);
+    uint4 b1 = philox10(seed, g1, offset);
+    uint32_t w[8] = {b0.x, b0.y, b0.z, b0.w, b1.x, b1.y, b1.z, b1.w};
+    float r[8], sn[8], cs[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float u1 = fmaxf((float)(w[j] & 0xffffu) * (1.0f / 65536.0f),
+                       1.1754944e-38f);
+      r[j] = sqrtf(-1.3862943611f * __log2f(u1));
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      __sincosf(6.2831853071795865f * ((float)(w[j] >> 16) * (1.0f / 65536.0f)),
+                &sn[j], &cs[j]);
+    }
+    V8 v0, v1;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      v0.v[j * 2 + 0] = __float2bfloat16(fmaf(r[j] * cs[j], b, a));
+      v0.v[j * 2 + 1] = __float2bfloat16(fmaf(r[j] * sn[j], b, a));
+      v1.v[j * 2 + 0] = __float2bfloat16(fmaf(r[4 + j] * cs[4 + j], b, a));
+      v1.v[j * 2 + 1] = __float2bfloat16(fmaf(r[4 + j] * sn[4 + j], b, a));
+    }
+    *reinterpret_cast<V8*>(out + (uint64_t)g0 * 8) = v0;
+    if (g1 < n8) {
+      *reinterpret_cast<V8*>(out + (uint64_t)g1 * 8) = v1;
+    }
+  }
+}
+
+template <typename K>
+double benchB(K kernel, __hip_bfloat16* buf, uint32_t n8, int blocks,
+              int iters, int threads) {
+  hipEvent_t e0, e1;
+  hipEventCreate(&e0);
+  hipEventCreate(&e1);
+  kernel<<<blocks, threads>>>(buf, n8, -1.f, 2.f, 42, 4);  // warm
+  hipDeviceSynchronize();
+  hipEventRecord(e0);
+  for (int i = 0; i < iters; ++i) {
+    kernel<<<blocks, threads>>>(buf, n8, -1.f, 2.f, 42, 4);
+  }
+  hipEventRecord(e1);
+  hipEventSynchronize(e1);
+  float ms;
+  hipEventElapsedTime(&ms, e0, e1);
+  double bytes = (double)n8 * 16.0 * iters;
+  return bytes / (ms / 1e3) / 1e12;  // TB/s
+}
+
 template <typename K>
 double bench(K kernel, __hip_bfloat16* buf, uint32_t n8, int blocks,
              int iters) {
@@ -281,6 +341,10 @@ int main() {
     printf("  normal  1xphilox (v2): %.2f TB/s\n", bench(normal_bf16_u16, buf, n8, blocks, 5));
     printf("  normal  v2 ILPx2     : %.2f TB/s\n", bench(normal_bf16_u16x2, buf, n8, blocks, 5));
     printf("  normal  v2 log2      : %.2f TB/s\n", bench(normal_bf16_log2, buf, n8, blocks, 5));
+    printf("  normal  split2 t256  : %.2f TB/s\n", benchB(normal_bf16_split2, buf, n8, blocks, 5, 256));
+    printf("  normal  split2 t512  : %.2f TB/s\n", benchB(normal_bf16_split2, buf, n8, blocks/2, 5, 512));
+    printf("  normal  log2 t512    : %.2f TB/s\n", benchB(normal_bf16_log2, buf, n8, blocks/2, 5, 512));
+    printf("  normal  log2 t128    : %.2f TB/s\n", benchB(normal_bf16_log2, buf, n8, blocks*2, 5, 128));
   }
   // memset reference ceiling
   hipEvent_t e0, e1;
