@@ -347,3 +347,39 @@ def test_fp8_tensors_through_fake_and_deferred() -> None:
     w = materialize_tensor(cast(Tensor, m.w8))
     assert w.dtype == torch.float8_e4m3fn
     assert w.float().abs().sum().item() > 0
+
+
+def test_nested_deferred_init() -> None:
+    # A deferred_init inside a deferred_init records into one tape; the
+    # nesting is level-counted, not scoped per call.
+    class Inner(Module):
+        def __init__(self):
+            super().__init__()
+            self.p = Parameter(torch.full([3], 2.0))
+
+    class Outer(Module):
+        def __init__(self):
+            super().__init__()
+            self.inner = deferred_init(Inner)
+            self.q = Parameter(torch.full([3], 5.0))
+
+    m = deferred_init(Outer)
+    assert is_deferred(m.inner)
+    materialize_module(m)
+    assert torch.equal(m.inner.p.detach(), torch.full([3], 2.0))
+    assert torch.equal(m.q.detach(), torch.full([3], 5.0))
+
+
+def test_materialize_during_recording_via_item() -> None:
+    # A terminal op mid-recording materializes its inputs while the outer
+    # recording continues and stays consistent.
+    class M(Module):
+        def __init__(self):
+            super().__init__()
+            a = torch.full([2], 3.0)
+            s = a.sum().item()       # forces materialization of `a`'s chain
+            self.p = Parameter(a * s)
+
+    m = deferred_init(M)
+    p = materialize_tensor(cast(Tensor, m.p))
+    assert torch.equal(p.detach(), torch.full([2], 18.0))
