@@ -150,7 +150,13 @@ class AnyPrecisionAdamW(Optimizer):
                     if self.use_fused is not None
                     else _fused_step_available(p)
                 )
-                if fused_ok and _fused_step_available(p):
+                if (
+                    fused_ok
+                    and _fused_step_available(p)
+                    and p.is_contiguous()
+                    and exp_avg.is_contiguous()
+                    and exp_avg_sq.is_contiguous()
+                ):
                     from torchdistx_amd import _kernels
 
                     _kernels.anyprecision_adamw_(
