@@ -92,8 +92,8 @@ bool tryNativeInitRedirect(
     }
     const at::Tensor& src = stack[1].toTensor();
     auto sst = src.scalar_type();
-    if (!src.defined() || !src.is_cuda() || !src.is_contiguous() ||
-        !src.sizes().equals(self.sizes()) ||
+    if (!src.defined() || src.device() != self.device() ||
+        !src.is_contiguous() || !src.sizes().equals(self.sizes()) ||
         (sst != at::kFloat && sst != at::kBFloat16 && sst != at::kHalf)) {
       return false;
     }
@@ -112,6 +112,11 @@ bool tryNativeInitRedirect(
   const bool is_rng = std::strcmp(redirect->tdx_name, "uniform_") == 0 ||
                       std::strcmp(redirect->tdx_name, "normal_") == 0;
   if (is_rng) {
+    // An explicitly passed generator must be honored; only the default
+    // generator's stream is replaced by the pinned counter-based one.
+    if (stack.size() >= 4 && !stack[3].isNone()) {
+      return false;
+    }
     // tdx RNG schemas carry two trailing optional args: the Philox seed
     // and counter offset pinned at record time (partition-invariant init).
     if (philox.has_value()) {
