@@ -907,13 +907,14 @@ ChainStep classifyChainNode(const OpNode& node) {
   } else if (chainNameIs(op.name, "aten::full")) {
     step.kind = ChainStep::Kind::kFill;
     step.p0 = scalarArg(1);
-  } else if (op.name == "aten::uniform_") {
-    step.kind = ChainStep::Kind::kUniform;
-    step.p0 = scalarArg(1);
-    step.p1 = scalarArg(2);
-    step.philox = op.philox;
-  } else if (op.name == "aten::normal_") {
-    step.kind = ChainStep::Kind::kNormal;
+  } else if (op.name == "aten::uniform_" || op.name == "aten::normal_") {
+    TORCH_CHECK(op.args.size() < 4 || op.args[3].isNone(),
+                "slice materialization: `", op.name,
+                "` was recorded with an explicit generator, which the "
+                "counter-based shard path cannot honor; materialize the "
+                "tensor fully instead.");
+    step.kind = op.name == "aten::uniform_" ? ChainStep::Kind::kUniform
+                                            : ChainStep::Kind::kNormal;
     step.p0 = scalarArg(1);
     step.p1 = scalarArg(2);
     step.philox = op.philox;
