@@ -288,3 +288,30 @@ def test_dtensor_materialization_matches_full() -> None:
     # into DTensors; the assembled full tensors equal a native full
     # materialization bitwise.
     assert all(run_distributed(_dtensor_worker, 2))
+
+
+def test_slice_falls_back_for_explicit_generator() -> None:
+    # An RNG op recorded with an explicit generator cannot use the
+    # counter-based shard path; the wrapper transparently falls back to the
+    # (generator-honoring) full replay and slices.
+    from torchdistx_amd import _C, deferred_init
+    from torchdistx_amd.parallel import materialize_tensor_shard
+    from torch.nn import Module, Parameter
+
+    g = torch.Generator().manual_seed(3)
+
+    class M(Module):
+        def __init__(self):
+            super().__init__()
+            p = torch.empty(8, 4)
+            p.uniform_(0, 1, generator=g)
+            self.p = Parameter(p)
+
+    m = deferred_init(M)
+    with pytest.raises(RuntimeError, match="explicit generator"):
+        _C.materialize_tensor_shard(m.p, 0, 4)
+    shard = materialize_tensor_shard(m.p, 0, 4)
+
+    g2 = torch.Generator().manual_seed(3)
+    expected = torch.empty(8, 4).uniform_(0, 1, generator=g2)
+    assert torch.equal(shard.detach(), expected[:4])

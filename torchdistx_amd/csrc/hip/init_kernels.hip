@@ -8,9 +8,9 @@
 // tensor lives on the GPU (csrc/core/native_redirect.cc).
 //
 // Design (per the CDNA4 kernel playbook for memory-bound elementwise ops):
-//  * 256-thread blocks (4 wave64s), grid-stride loops capped at 2048
-//    blocks (256 CUs x 8 blocks) so the chip is filled without
-//    oversubscribing the launch queue;
+//  * 256-thread blocks (4 wave64s), grid-stride loops capped at 16384
+//    blocks (measured sweet spot for this ALU-heavy streaming shape;
+//    see kMaxBlocks below);
 //  * every store is 16 bytes per lane (float4 / 8 x bf16) — 1 KiB per wave
 //    per instruction, the HBM coalescing sweet spot; scalar tail only for
 //    the last partial group;
@@ -45,6 +45,8 @@ constexpr int kMaxBlocks = 16384;
 
 // ---------------------------------------------------------------------------
 // Philox4x32-10 (standard constants), producing 4 x uint32 per invocation.
+// Kept in sync with the CPU reference in csrc/core/philox.h (identical
+// constants and integer pipeline).
 // ---------------------------------------------------------------------------
 
 __device__ __forceinline__ uint2 mulhilo32(uint32_t a, uint32_t b) {
@@ -116,8 +118,9 @@ __device__ __forceinline__ T from_float(float v) {
   }
 }
 
-// Elements per thread-group-iteration: 16-bit types pack 8 elements into one
-// 16-byte store (two Philox calls); 32-bit types pack 4 (one call).
+// Elements per thread-group-iteration: 16-bit types pack 8 elements into
+// one 16-byte store; 32-bit types pack 4 — one philox10 per store either
+// way (16-bit dtypes draw 16 random bits per sample).
 template <typename T>
 struct VecTraits;
 template <>
