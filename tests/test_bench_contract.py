@@ -1,0 +1,46 @@
+# Guards the driver contract of bench.py: one JSON line on stdout with the
+# agreed fields, runnable without flags on a GPU-less machine.
+
+import json
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _run(*extra):
+    result = subprocess.run(
+        [sys.executable, "bench.py", "--model", "tiny", "--steps", "2",
+         "--warmup", "1", *extra],
+        capture_output=True,
+        text=True,
+        timeout=300,
+        cwd=REPO,
+    )
+    assert result.returncode == 0, result.stderr
+    lines = [l for l in result.stdout.strip().splitlines() if l.strip()]
+    assert len(lines) == 1, f"expected exactly one JSON line: {lines}"
+    return json.loads(lines[0])
+
+
+def test_bench_json_contract() -> None:
+    d = _run()
+    for key in (
+        "metric", "value", "unit", "n_gpus", "steps", "warmup",
+        "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+        "dtype", "data", "config",
+    ):
+        assert key in d, key
+    assert d["n_gpus"] == 1
+    assert d["steps"] == 2 and d["warmup"] == 1
+    assert d["higher_is_better"] is True
+    assert d["scaling"] in ("weak", "strong")
+    assert d["value"] > 0 and d["ms_per_step"] > 0
+    assert d["config"]["model"] == "tiny"
+    assert "synthetic" in d["data"]
+
+
+def test_bench_modes_report_scaling() -> None:
+    assert _run("--mode", "replicate")["scaling"] == "weak"
+    assert _run("--mode", "slice")["scaling"] == "strong"
