@@ -383,3 +383,13 @@ def test_materialize_during_recording_via_item() -> None:
     m = deferred_init(M)
     p = materialize_tensor(cast(Tensor, m.p))
     assert torch.equal(p.detach(), torch.full([2], 18.0))
+
+
+def test_saving_deferred_module_raises_loudly() -> None:
+    # Fake tensors have no storage; serializing a deferred module fails
+    # loudly instead of writing garbage (materialize first).
+    import io
+
+    m = deferred_init(torch.nn.Linear, 3, 3)
+    with pytest.raises((NotImplementedError, RuntimeError)):
+        torch.save(m.state_dict(), io.BytesIO())
