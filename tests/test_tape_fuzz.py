@@ -56,14 +56,14 @@ def _build_program(rng: random.Random, n_steps: int):
     return ops
 
 
-def _run_program(ops, device="cpu"):
+def _run_program(ops, device="cpu", dtype=torch.float32):
     ts = []
     fam = []  # structural alias family per tensor (identical for fake/real)
     for op in ops:
         kind = op[0]
         if kind == "new":
             _, n, val = op
-            ts.append(torch.full([n, n], val, device=device))
+            ts.append(torch.full([n, n], val, device=device, dtype=dtype))
             fam.append(len(fam))
         elif kind == "view":
             base = ts[op[1]]
@@ -126,16 +126,16 @@ def _run_program(ops, device="cpu"):
     return ts
 
 
-def _check_seed(seed, device):
+def _check_seed(seed, device, dtype=torch.float32):
     rng = random.Random(seed)
     ops = _build_program(rng, n_steps=25)
 
-    eager = _run_program(ops, device)
+    eager = _run_program(ops, device, dtype)
 
     class Holder(Module):
         def __init__(self):
             super().__init__()
-            self.tensors = _run_program(ops, device)
+            self.tensors = _run_program(ops, device, dtype)
 
     holder = deferred_init(Holder)
 
@@ -149,6 +149,12 @@ def _check_seed(seed, device):
 @pytest.mark.parametrize("seed", range(30))
 def test_random_program_replay_matches_eager(seed) -> None:
     _check_seed(seed, "cpu")
+
+
+@pytest.mark.parametrize("seed", range(10))
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float16])
+def test_random_program_replay_low_precision(seed, dtype) -> None:
+    _check_seed(seed + 40_000, "cpu", dtype)
 
 
 @pytest.mark.gpu
