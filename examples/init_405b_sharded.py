@@ -26,6 +26,12 @@ from torchdistx_amd.utils import describe_module
 
 
 def main():
+    if "RANK" not in os.environ:  # plain `python ...` -> single-rank run
+        os.environ.setdefault("RANK", "0")
+        os.environ.setdefault("LOCAL_RANK", "0")
+        os.environ.setdefault("WORLD_SIZE", "1")
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29572")
     dist.init_process_group("nccl")
     rank = dist.get_rank()
     world = dist.get_world_size()

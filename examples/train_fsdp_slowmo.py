@@ -28,6 +28,12 @@ from torchdistx_amd.slowmo import SlowMomentumOptimizer, SlowMoState, slowmo_hoo
 
 
 def main():
+    if "RANK" not in os.environ:  # plain `python ...` -> single-rank run
+        os.environ.setdefault("RANK", "0")
+        os.environ.setdefault("LOCAL_RANK", "0")
+        os.environ.setdefault("WORLD_SIZE", "1")
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29571")
     dist.init_process_group("nccl")
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     torch.cuda.set_device(local_rank)
