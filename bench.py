@@ -89,6 +89,23 @@ def main():
     dtype = {"bf16": torch.bfloat16, "fp32": torch.float32,
              "fp16": torch.float16}[args.dtype]
 
+    if use_cuda:
+        # Fail fast with a clear message instead of an allocator OOM deep
+        # inside materialization (e.g. llama3-405b needs >1 rank for any
+        # non-replicated mode; replicate of it fits nowhere).
+        per_rank = cfg.n_params * dtype.itemsize
+        if args.mode in ("shard", "slice"):
+            per_rank = -(-per_rank // world)  # this rank's share (ceil)
+        hbm = torch.cuda.get_device_properties(0).total_memory
+        if per_rank > hbm:  # Mixtral's 281 GB of 288 GB is fine; 812 GB is not
+            print(
+                f"FATAL: {model_name} mode={args.mode} needs "
+                f"{per_rank / 1e9:.0f} GB per rank but the device has "
+                f"{hbm / 1e9:.0f} GB; use more ranks or mode=slice/shard",
+                file=sys.stderr,
+            )
+            sys.exit(1)
+
     def one_step(seed: int) -> None:
         torch.manual_seed(seed)
         module = deferred_init(build_model, cfg, device=device, dtype=dtype)
