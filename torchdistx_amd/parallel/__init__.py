@@ -1,3 +1,6 @@
+from torchdistx_amd.parallel.fsdp2 import (  # noqa: F401
+    fully_shard_deferred,
+)
 from torchdistx_amd.parallel.sharded_materialize import (  # noqa: F401
     assign_owners,
     materialize_experts_sharded,
