@@ -32,11 +32,15 @@ def single_proc_group():
 
 
 def test_fully_shard_deferred_single_rank(single_proc_group) -> None:
+    # Pin the mesh to CPU: on a GPU box fully_shard's default mesh would
+    # otherwise target cuda while this test's tensors live on CPU.
+    from torch.distributed.device_mesh import init_device_mesh
     from torch.distributed.tensor import DTensor
 
+    mesh = init_device_mesh("cpu", (1,))
     m = deferred_init(_make)
     assert is_deferred(m)
-    fully_shard_deferred(m, submodules=[m[0], m[2]])
+    fully_shard_deferred(m, submodules=[m[0], m[2]], mesh=mesh)
     assert not is_deferred(m)
     for p in m.parameters():
         assert isinstance(p.data, DTensor)
@@ -51,9 +55,12 @@ def test_fully_shard_deferred_single_rank(single_proc_group) -> None:
 
 
 def _fsdp2_worker(rank, world_size):
+    from torch.distributed.device_mesh import init_device_mesh
+
+    mesh = init_device_mesh("cpu", (world_size,))
     torch.manual_seed(rank)  # rank-skewed; deferred tape must override
     m = deferred_init(_make)
-    fully_shard_deferred(m, submodules=[m[0], m[2]])
+    fully_shard_deferred(m, submodules=[m[0], m[2]], mesh=mesh)
 
     torch.manual_seed(99)
     x = torch.randn(3, 16)  # same batch on every rank
