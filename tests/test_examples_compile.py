@@ -13,7 +13,12 @@ _EXAMPLES = os.path.join(
 
 
 @pytest.mark.parametrize(
-    "name", ["train_fsdp_slowmo.py", "init_405b_sharded.py"]
+    "name",
+    [
+        "train_fsdp_slowmo.py",
+        "train_fsdp2_anyprecision.py",
+        "init_405b_sharded.py",
+    ],
 )
 def test_example_compiles(name: str) -> None:
     py_compile.compile(os.path.join(_EXAMPLES, name), doraise=True)
