@@ -140,3 +140,30 @@ def test_fsdp2_fully_shard_deferred(nccl_world1) -> None:
     xr = torch.randint(0, TINY.vocab_size, (2, 16), device="cuda")
     assert torch.equal(out.detach(), ref(xr).detach())
     torch.cuda.synchronize()
+
+
+def test_fsdp2_anyprecision_optimizer_step(nccl_world1) -> None:
+    # Regression: FSDP2 DTensor parameters are storage-less wrapper
+    # subclasses; the fused HIP AdamW step must refuse them (null device
+    # pointer) and take the eager op sequence, which dispatches through
+    # DTensor correctly.
+    from torchdistx_amd import deferred_init
+    from torchdistx_amd.models import TINY, build_model
+    from torchdistx_amd.optimizers import AnyPrecisionAdamW
+    from torchdistx_amd.parallel import fully_shard_deferred
+
+    torch.manual_seed(0)
+    module = deferred_init(build_model, TINY, device="cuda",
+                           dtype=torch.bfloat16)
+    fully_shard_deferred(module, submodules=list(module.blocks))
+    optim = AnyPrecisionAdamW(
+        module.parameters(), lr=1e-3, use_kahan_summation=True
+    )
+    for step in range(3):
+        x = torch.randint(0, TINY.vocab_size, (2, 16), device="cuda")
+        loss = module(x).float().square().mean()
+        optim.zero_grad()
+        loss.backward()
+        optim.step()
+    torch.cuda.synchronize()
+    assert torch.isfinite(loss.detach()).item()

@@ -21,6 +21,12 @@ from torch.optim.optimizer import Optimizer
 def _fused_step_available(p: torch.Tensor) -> bool:
     if p.device.type != "cuda":
         return False
+    # Tensor subclasses (DTensor under FSDP2, functorch wrappers, ...) are
+    # storage-less wrapper tensors: handing them to the raw HIP kernel
+    # would dereference a null device pointer. They take the eager op
+    # sequence, which dispatches through the subclass correctly.
+    if torch.utils._python_dispatch.is_traceable_wrapper_subclass(p):
+        return False
     try:
         from torchdistx_amd import _kernels
 
