@@ -339,6 +339,13 @@ void anyprecision_adamw_step(at::Tensor& param,
               "fused AnyPrecisionAdamW requires GPU tensors");
   const at::Tensor* checked[] = {&param, &grad, &exp_avg, &exp_avg_sq};
   for (const at::Tensor* t : checked) {
+    // Wrapper subclasses (DTensor, ...) reach C++ as storage-less impls;
+    // dereferencing their null device pointer would be a GPU memory
+    // fault, so refuse them here rather than in the kernel.
+    TORCH_CHECK(t->unsafeGetTensorImpl()->has_storage() &&
+                    t->const_data_ptr() != nullptr,
+                "fused AnyPrecisionAdamW requires plain dense tensors "
+                "(got a storage-less tensor, e.g. a DTensor wrapper)");
     TORCH_CHECK(t->is_contiguous(),
                 "fused AnyPrecisionAdamW requires contiguous tensors");
     TORCH_CHECK(t->numel() == param.numel(),
