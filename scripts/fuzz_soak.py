@@ -1,6 +1,6 @@
 """Extended differential fuzz soak of the deferred-init tape (the CI test
 runs 30 seeds; this runs thousands). Usage: python scripts/fuzz_soak.py
-[n_seeds] [device]."""
+[n_seeds] [device] [seed_base]."""
 import os
 import random
 import sys
@@ -17,9 +17,10 @@ from torchdistx_amd.deferred_init import deferred_init, materialize_tensor
 
 n_seeds = int(sys.argv[1]) if len(sys.argv) > 1 else 1000
 device = sys.argv[2] if len(sys.argv) > 2 else "cpu"
+seed_base = int(sys.argv[3]) if len(sys.argv) > 3 else 0
 
 fails = 0
-for seed in range(n_seeds):
+for seed in range(seed_base, seed_base + n_seeds):
     rng = random.Random(seed * 7919 + 13)
     ops = _build_program(rng, 45)
     eager = _run_program(ops, device)
@@ -39,7 +40,7 @@ for seed in range(n_seeds):
             print("FAIL", seed, i)
             break
     if seed % 500 == 499:
-        print(f"{seed + 1}/{n_seeds} seeds, {fails} fails")
+        print(f"{seed + 1 - seed_base}/{n_seeds} seeds, {fails} fails")
 
 print(f"done: {fails} fails over {n_seeds} seeds on {device}")
 sys.exit(1 if fails else 0)
