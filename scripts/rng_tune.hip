@@ -42,6 +42,28 @@ __device__ __forceinline__ uint4 philox10(uint64_t seed, uint64_t subseq,
   return c;
 }
 
+// Round-count-parametrized Philox (7 rounds passes BigCrush per Salmon
+// et al. 2011, Table 2; 10 is PyTorch's margin choice). Measures how much
+// of the normal-kernel plateau is Philox VALU work.
+template <int R>
+__device__ __forceinline__ uint4 philoxR(uint64_t seed, uint64_t subseq,
+                                         uint64_t offset) {
+  constexpr uint32_t kW0 = 0x9E3779B9u, kW1 = 0xBB67AE85u;
+  constexpr uint32_t kM0 = 0xD2511F53u, kM1 = 0xCD9E8D57u;
+  uint32_t k0 = (uint32_t)seed, k1 = (uint32_t)(seed >> 32);
+  uint4 c = make_uint4((uint32_t)offset, (uint32_t)(offset >> 32),
+                       (uint32_t)subseq, (uint32_t)(subseq >> 32));
+#pragma unroll
+  for (int r = 0; r < R; ++r) {
+    uint2 r0 = mulhilo32(kM0, c.x);
+    uint2 r1 = mulhilo32(kM1, c.z);
+    c = make_uint4(r1.y ^ c.y ^ k0, r1.x, r0.y ^ c.w ^ k1, r0.x);
+    k0 += kW0;
+    k1 += kW1;
+  }
+  return c;
+}
+
 __device__ __forceinline__ float u2f(uint32_t x) {
   return (float)(x >> 8) * (1.0f / 16777216.0f);
 }
@@ -136,6 +158,49 @@ __global__ void normal_bf16_u16(__hip_bfloat16* __restrict__ out, uint32_t n8,
                      (float)(w[j] >> 16) * (1.0f / 65536.0f));
       v.v[j * 2 + 0] = __float2bfloat16(fmaf(nj.x, b, a));
       v.v[j * 2 + 1] = __float2bfloat16(fmaf(nj.y, b, a));
+    }
+    *reinterpret_cast<V8*>(out + (uint64_t)g * 8) = v;
+  }
+}
+
+// Philox-round sweep of the shipped v2-log2 shape.
+template <int R>
+__global__ void normal_bf16_rounds(__hip_bfloat16* __restrict__ out,
+                                   uint32_t n8, float a, float b,
+                                   uint64_t seed, uint64_t offset) {
+  const uint32_t stride = gridDim.x * blockDim.x;
+  for (uint32_t g = blockIdx.x * blockDim.x + threadIdx.x; g < n8;
+       g += stride) {
+    uint4 bits = philoxR<R>(seed, g, offset);
+    uint32_t w[4] = {bits.x, bits.y, bits.z, bits.w};
+    V8 v;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float2 nj = bm((float)(w[j] & 0xffffu) * (1.0f / 65536.0f),
+                     (float)(w[j] >> 16) * (1.0f / 65536.0f));
+      v.v[j * 2 + 0] = __float2bfloat16(fmaf(nj.x, b, a));
+      v.v[j * 2 + 1] = __float2bfloat16(fmaf(nj.y, b, a));
+    }
+    *reinterpret_cast<V8*>(out + (uint64_t)g * 8) = v;
+  }
+}
+
+template <int R>
+__global__ void uniform_bf16_rounds(__hip_bfloat16* __restrict__ out,
+                                    uint32_t n8, float a, float b,
+                                    uint64_t seed, uint64_t offset) {
+  const uint32_t stride = gridDim.x * blockDim.x;
+  for (uint32_t g = blockIdx.x * blockDim.x + threadIdx.x; g < n8;
+       g += stride) {
+    uint4 bits = philoxR<R>(seed, g, offset);
+    uint32_t w[4] = {bits.x, bits.y, bits.z, bits.w};
+    V8 v;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      v.v[j * 2 + 0] = __float2bfloat16(
+          fmaf((float)(w[j] & 0xffffu) * (1.0f / 65536.0f), b, a));
+      v.v[j * 2 + 1] = __float2bfloat16(
+          fmaf((float)(w[j] >> 16) * (1.0f / 65536.0f), b, a));
     }
     *reinterpret_cast<V8*>(out + (uint64_t)g * 8) = v;
   }
@@ -345,6 +410,10 @@ int main() {
     printf("  normal  split2 t512  : %.2f TB/s\n", benchB(normal_bf16_split2, buf, n8, blocks/2, 5, 512));
     printf("  normal  log2 t512    : %.2f TB/s\n", benchB(normal_bf16_log2, buf, n8, blocks/2, 5, 512));
     printf("  normal  log2 t128    : %.2f TB/s\n", benchB(normal_bf16_log2, buf, n8, blocks*2, 5, 128));
+    printf("  normal  rounds=10    : %.2f TB/s\n", bench(normal_bf16_rounds<10>, buf, n8, blocks, 5));
+    printf("  normal  rounds=7     : %.2f TB/s\n", bench(normal_bf16_rounds<7>, buf, n8, blocks, 5));
+    printf("  uniform rounds=10    : %.2f TB/s\n", bench(uniform_bf16_rounds<10>, buf, n8, blocks, 5));
+    printf("  uniform rounds=7     : %.2f TB/s\n", bench(uniform_bf16_rounds<7>, buf, n8, blocks, 5));
   }
   // memset reference ceiling
   hipEvent_t e0, e1;
