@@ -6,6 +6,8 @@ import os
 import subprocess
 import sys
 
+import pytest
+
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
@@ -44,3 +46,17 @@ def test_bench_json_contract() -> None:
 def test_bench_modes_report_scaling() -> None:
     assert _run("--mode", "replicate")["scaling"] == "weak"
     assert _run("--mode", "slice")["scaling"] == "strong"
+
+
+@pytest.mark.gpu
+def test_bench_json_contract_gpu() -> None:
+    # Same contract, exercised on the GPU path (native kernels required,
+    # device-memory guard, cuda synchronize bracketing).
+    import torch
+
+    if not torch.cuda.is_available():
+        pytest.skip("needs a ROCm GPU")
+    d = _run()
+    assert d["config"]["device"] == "cuda"
+    assert d["config"]["native_init_kernels"] is True
+    assert d["value"] > 0
