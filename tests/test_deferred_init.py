@@ -393,3 +393,99 @@ def test_saving_deferred_module_raises_loudly() -> None:
     m = deferred_init(torch.nn.Linear, 3, 3)
     with pytest.raises((NotImplementedError, RuntimeError)):
         torch.save(m.state_dict(), io.BytesIO())
+
+
+# ---------------------------------------------------------------------------
+# RNG session semantics (see docs/deferred_init.md "RNG contract").
+# ---------------------------------------------------------------------------
+
+
+def _native_cpu_init(model_fn):
+    """Records and materializes through the pinned-Philox CPU-native path."""
+    from torchdistx_amd import _C
+
+    m = deferred_init(model_fn)
+    _C.set_native_init_cpu(True)
+    try:
+        materialize_module(m)
+    finally:
+        _C.set_native_init_cpu(False)
+    return m
+
+
+def test_unseeded_sessions_draw_independent_native_streams() -> None:
+    # Two deferred_init sessions without re-seeding must produce different
+    # weights on the pinned (native) path, like eager code would.
+    torch.manual_seed(1000)
+    m1 = _native_cpu_init(lambda: torch.nn.Linear(16, 16))
+    m2 = _native_cpu_init(lambda: torch.nn.Linear(16, 16))
+    assert not torch.equal(m1.weight, m2.weight)
+    assert not torch.equal(m1.bias, m2.bias)
+
+
+def test_reseeded_sessions_pin_identical_native_streams() -> None:
+    # Same seed -> bitwise-identical native init (pure function of the
+    # generator state at session entry).
+    torch.manual_seed(1001)
+    m1 = _native_cpu_init(lambda: torch.nn.Linear(16, 16))
+    torch.manual_seed(1001)
+    m2 = _native_cpu_init(lambda: torch.nn.Linear(16, 16))
+    assert torch.equal(m1.weight, m2.weight)
+    assert torch.equal(m1.bias, m2.bias)
+
+
+def test_unseeded_sessions_differ_on_stock_cpu_replay() -> None:
+    # Also holds for the stock-ATen CPU replay path: each session snapshots
+    # a different generator state into its replay cursor.
+    torch.manual_seed(1002)
+    m1 = deferred_init(torch.nn.Linear, 16, 16)
+    m2 = deferred_init(torch.nn.Linear, 16, 16)
+    materialize_module(m1)
+    materialize_module(m2)
+    assert not torch.equal(m1.weight, m2.weight)
+
+
+def test_cpu_replay_does_not_consume_ambient_generator() -> None:
+    # Replay draws from the session's snapshot cursor, not the live
+    # generator: materialization leaves the ambient RNG stream untouched.
+    torch.manual_seed(1003)
+    m = deferred_init(torch.nn.Linear, 8, 8)
+    before = torch.get_rng_state()
+    materialize_module(m)
+    after = torch.get_rng_state()
+    assert torch.equal(before, after)
+
+
+def test_concurrent_thread_sessions_are_independent() -> None:
+    # Sessions are thread-local: two threads recording at once must not
+    # clobber each other's seed/slot state into colliding pins.
+    import threading
+
+    torch.manual_seed(1004)
+    results = {}
+    barrier = threading.Barrier(2)
+
+    def build(key):
+        barrier.wait()
+        results[key] = _native_cpu_init(lambda: torch.nn.Linear(16, 16))
+
+    threads = [
+        threading.Thread(target=build, args=(i,)) for i in range(2)
+    ]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert not torch.equal(results[0].weight, results[1].weight)
+
+
+def test_use_fused_true_raises_when_unavailable() -> None:
+    # use_fused=True must never silently fall back to the eager op
+    # sequence (a benchmark would silently measure the wrong path).
+    from torchdistx_amd.optimizers import AnyPrecisionAdamW
+
+    p = torch.nn.Parameter(torch.randn(8))
+    p.grad = torch.randn(8)
+    opt = AnyPrecisionAdamW([p], lr=1e-3, use_fused=True)
+    with pytest.raises(RuntimeError, match="use_fused=True"):
+        opt.step()

@@ -27,6 +27,7 @@
 #include <unordered_set>
 #include <vector>
 
+#include <ATen/CPUGeneratorImpl.h>
 #include <ATen/Context.h>
 #include <ATen/ThreadLocalState.h>
 #include <ATen/core/dispatch/Dispatcher.h>
@@ -94,26 +95,42 @@ struct RecordedOp {
 // the native CDNA4 kernels run the replay).
 constexpr uint64_t kPhiloxStridePerOp = 4;
 
-// Session state: the seed is a 64-bit nonce derived (by hashing, WITHOUT
-// consuming) from the default CPU generator's state when the outermost
-// deferred_init is entered, and slots count from 1 within the session.
-// Consequences (all intentional):
-//   * the pinned init bits are a pure function of the RNG state at
-//     deferred_init entry — torch.manual_seed(S) fixes them, exactly like
-//     eager init is a function of the generator state at construction;
-//   * the generator stream is never perturbed, so CPU replay (which uses
-//     the stock generator) stays bitwise-equal to eager construction;
-//   * within one tape, replaying any subset on any rank gives identical
-//     bits (partition-invariant sharded materialization).
-std::atomic<uint64_t> session_rng_slot{1};
-std::atomic<uint64_t> session_rng_seed{0};
+// RNG session, one per outermost deferred_init on each thread (sessions
+// are per-thread by construction: deferred_level is thread_local).
+//
+//   * `pin_seed` is a 64-bit nonce hashed from the FULL default CPU
+//     generator state at session entry. Pins are a pure function of that
+//     state, so torch.manual_seed(S) fixes the native init bits, exactly
+//     like eager init is a function of the generator state at
+//     construction — and within one tape, replaying any subset on any
+//     rank gives identical bits (partition-invariant sharded
+//     materialization).
+//   * Session entry then ADVANCES the live CPU generator by one 64-bit
+//     draw, the way eager construction consumes it: a second session
+//     without re-seeding hashes a different state and therefore draws an
+//     independent stream (two unseeded models never collide).
+//   * `replay_cursor` snapshots the pre-advance state. CPU stock replay
+//     of the session's RNG ops draws from this cursor (not the ambient
+//     generator), so a full in-tape-order CPU materialization is
+//     bitwise-equal to eager construction under the same seed, and replay
+//     never perturbs the ambient generator stream.
+struct RngSession {
+  uint64_t pin_seed = 0;
+  uint64_t next_slot = 1;
+  // Default CPU generator state; advanced only by this session's CPU
+  // stock replay, in tape order. Guarded by tape_mutex (replay holds it).
+  at::Tensor replay_cursor;
+};
+
+thread_local std::shared_ptr<RngSession> current_rng_session;
 
 bool isRngOpName(const std::string& name) {
   return name == "aten::uniform_" || name == "aten::normal_";
 }
 
 void beginRngSession() {
-  // FNV-1a over the seed and a prefix of the generator state: equal state
+  auto session = std::make_shared<RngSession>();
+  // FNV-1a over the seed and the full generator state: equal state
   // -> equal nonce; any consumption or re-seeding -> different nonce.
   auto gen = at::globalContext().defaultGenerator(c10::DeviceType::CPU);
   uint64_t h = 1469598103934665603ull;
@@ -127,7 +144,7 @@ void beginRngSession() {
     mix(gen.current_seed());
     at::Tensor state = gen.get_state();
     const auto* bytes = state.const_data_ptr<uint8_t>();
-    int64_t take = std::min<int64_t>(state.numel(), 64);
+    const int64_t take = state.numel();
     uint64_t word = 0;
     for (int64_t i = 0; i < take; ++i) {
       word = (word << 8) | bytes[i];
@@ -136,23 +153,29 @@ void beginRngSession() {
         word = 0;
       }
     }
+    if ((take & 7) != 0) {
+      mix(word);
+    }
+    session->replay_cursor = std::move(state);  // get_state returns a copy
+    // Consume one draw so the next unseeded session sees a new state.
+    gen.get<at::CPUGeneratorImpl>()->random64();
   }
-  session_rng_seed.store(h, std::memory_order_relaxed);
-  session_rng_slot.store(1, std::memory_order_relaxed);
+  session->pin_seed = h;
+  current_rng_session = std::move(session);
 }
 
 // Pins (seed, counter-offset) for every recorded RNG op. The pins drive
 // the tdx Philox kernels on GPU replay and the slice-materialization fast
-// path on any device; plain CPU replay ignores them (it uses the stock
-// generator for eager bit-parity).
+// path on any device; plain CPU replay ignores them (it draws from the
+// session's replay cursor for eager bit-parity).
 std::optional<std::pair<uint64_t, uint64_t>> pinPhiloxForOp(
     const std::string& name) {
-  if (!isRngOpName(name)) {
+  if (!isRngOpName(name) || current_rng_session == nullptr) {
     return std::nullopt;
   }
-  uint64_t seed = session_rng_seed.load(std::memory_order_relaxed);
-  uint64_t slot = session_rng_slot.fetch_add(1, std::memory_order_relaxed);
-  return std::make_pair(seed, slot * kPhiloxStridePerOp);
+  RngSession& session = *current_rng_session;
+  return std::make_pair(session.pin_seed,
+                        session.next_slot++ * kPhiloxStridePerOp);
 }
 
 struct InputSlot {
@@ -426,15 +449,41 @@ void deferredInitHandler(const c10::OperatorHandle& op,
 
   if (has_fake_arg || has_fake_ret) {
     auto philox = pinPhiloxForOp(schema.operator_name().name);
+    // RNG ops additionally carry their session, for the CPU replay cursor.
+    std::shared_ptr<RngSession> session =
+        philox.has_value() ? current_rng_session : nullptr;
     recordOp(
         schema.operator_name().name,
-        [handle = op, philox](torch::jit::Stack& s) {
+        [handle = op, philox, session](torch::jit::Stack& s) {
           // Replay hot path: recorded init ops whose target lives on the
           // GPU run through the hand-written CDNA4 kernels (tdx::) when
           // the _K extension is loaded. The pinned Philox state makes the
           // result independent of which rank replays which subset.
           if (tryNativeInitRedirect(handle, s, philox)) {
             return;
+          }
+          if (session != nullptr && !s.empty() && s.front().isTensor()) {
+            // CPU stock replay of a session RNG op: draw from the
+            // session's replay cursor instead of the ambient generator,
+            // so in-tape-order materialization reproduces eager bits and
+            // the ambient stream is left untouched. Ops recorded with an
+            // explicit generator keep consuming that generator.
+            const at::Tensor& self = s.front().toTensor();
+            const bool explicit_gen = s.size() >= 4 && !s[3].isNone();
+            if (self.defined() && self.is_cpu() && !explicit_gen) {
+              auto gen =
+                  at::globalContext().defaultGenerator(c10::DeviceType::CPU);
+              auto swap_state = [&gen](const at::Tensor& st) {
+                std::lock_guard<std::mutex> lock(gen.mutex());
+                at::Tensor prev = gen.get_state();
+                gen.set_state(st);
+                return prev;
+              };
+              at::Tensor ambient = swap_state(session->replay_cursor);
+              handle.callBoxed(s);
+              session->replay_cursor = swap_state(ambient);
+              return;
+            }
           }
           handle.callBoxed(s);
         },

@@ -26,6 +26,11 @@ void visitIValue(const c10::IValue& v, const TensorVisitor& visitor) {
     for (const c10::IValue& e : v.toTupleRef().elements()) {
       visitIValue(e, visitor);
     }
+  } else if (v.isGenericDict()) {
+    for (const auto& kv : v.toGenericDict()) {
+      visitIValue(kv.key(), visitor);
+      visitIValue(kv.value(), visitor);
+    }
   }
 }
 
@@ -67,6 +72,16 @@ c10::IValue mapIValue(const c10::IValue& v, const TensorMapFn& fn) {
     }
     if (changed) {
       return c10::ivalue::Tuple::create(std::move(elems));
+    }
+    return v;
+  }
+  if (v.isGenericDict()) {
+    auto dict = v.toGenericDict();
+    for (const auto& kv : dict) {
+      c10::IValue mapped = mapIValue(kv.value(), fn);
+      if (!mapped.isSameIdentity(kv.value())) {
+        dict.insert_or_assign(kv.key(), std::move(mapped));
+      }
     }
     return v;
   }

@@ -382,6 +382,10 @@ __global__ void rng_shard_kernel(T* __restrict__ out,
   const int64_t g_first = start / kElems;
   const int64_t g_last = (end + kElems - 1) / kElems;
   const int64_t stride = static_cast<int64_t>(gridDim.x) * blockDim.x;
+  // The vector path stores Vec packs at out + (base - start); base is a
+  // multiple of kElems, so the destination is 16-byte aligned only when
+  // `start` is too. Unaligned shards (odd row sizes) store elementwise.
+  const bool vec_aligned = (start % kElems) == 0;
 
   for (int64_t g = g_first +
            blockIdx.x * static_cast<int64_t>(blockDim.x) + threadIdx.x;
@@ -390,7 +394,7 @@ __global__ void rng_shard_kernel(T* __restrict__ out,
     rngGroupValues<T, kDist>(static_cast<uint64_t>(g), a, b, seed, offset,
                              vals);
     const int64_t base = g * kElems;
-    if (base >= start && base + kElems <= end) {
+    if (vec_aligned && base >= start && base + kElems <= end) {
       Vec v;
       T* vp = reinterpret_cast<T*>(&v);
 #pragma unroll

@@ -18,21 +18,42 @@ import torch
 from torch.optim.optimizer import Optimizer
 
 
-def _fused_step_available(p: torch.Tensor) -> bool:
-    if p.device.type != "cuda":
-        return False
-    # Tensor subclasses (DTensor under FSDP2, functorch wrappers, ...) are
-    # storage-less wrapper tensors: handing them to the raw HIP kernel
-    # would dereference a null device pointer. They take the eager op
-    # sequence, which dispatches through the subclass correctly.
-    if torch.utils._python_dispatch.is_traceable_wrapper_subclass(p):
-        return False
+def _local_view(t):
+    """DTensor (FSDP2) parameters carry their data in a plain contiguous
+    local shard; the fused kernel runs on that shard directly — an
+    elementwise optimizer step on the local shard is exactly the sharded
+    step. Other tensors pass through unchanged."""
+    if t is None:
+        return None
+    try:
+        from torch.distributed.tensor import DTensor
+    except ImportError:
+        return t
+    if isinstance(t, DTensor):
+        return t.to_local()
+    return t
+
+
+def _kernel_loaded() -> bool:
     try:
         from torchdistx_amd import _kernels
 
         return _kernels.has_anyprecision_adamw()
     except Exception:
         return False
+
+
+def _fused_step_available(p: torch.Tensor) -> bool:
+    p = _local_view(p)
+    if p.device.type != "cuda":
+        return False
+    # Non-DTensor wrapper subclasses (functorch wrappers, fake modes, ...)
+    # are storage-less: handing them to the raw HIP kernel would
+    # dereference a null device pointer. They take the eager op sequence,
+    # which dispatches through the subclass correctly.
+    if torch.utils._python_dispatch.is_traceable_wrapper_subclass(p):
+        return False
+    return _kernel_loaded()
 
 
 class AnyPrecisionAdamW(Optimizer):
@@ -52,11 +73,15 @@ class AnyPrecisionAdamW(Optimizer):
         variance_dtype: dtype of ``exp_avg_sq`` (default: torch.bfloat16).
         compensation_buffer_dtype: dtype of the Kahan compensation buffer
             (default: torch.bfloat16).
-        use_fused: force (True) or forbid (False) the fused single-kernel
-            CDNA4 step; None (default) uses it automatically on GPU
-            tensors when the kernel extension is loaded. The fused step
-            computes in fp32 and rounds once per state store, so it is
-            close to but not bitwise-identical with the eager op sequence.
+        use_fused: True forces the fused single-kernel CDNA4 step and
+            raises ``RuntimeError`` if it cannot run (so a benchmark can
+            never silently measure the eager path); False forbids it;
+            None (default) uses it automatically on GPU tensors when the
+            kernel extension is loaded. DTensor (FSDP2) parameters are
+            unwrapped to their local shard for the fused step. The fused
+            step computes in fp32 and rounds once per state store, so it
+            is close to but not bitwise-identical with the eager op
+            sequence.
     """
 
     def __init__(
@@ -84,6 +109,10 @@ class AnyPrecisionAdamW(Optimizer):
         )
         super().__init__(params, defaults)
         self.use_fused = use_fused
+        # Diagnostic: number of parameter updates that took the fused
+        # CDNA4 kernel (lets tests and benchmarks assert the fast path
+        # actually ran).
+        self._fused_steps = 0
 
     def load_state_dict(self, state_dict):
         """Base-class restore, then re-cast the optimizer states to their
@@ -151,26 +180,46 @@ class AnyPrecisionAdamW(Optimizer):
                 bias_correction2_sqrt = (1 - beta2**step) ** 0.5
                 step_size = lr / bias_correction1
 
-                fused_ok = (
-                    self.use_fused
-                    if self.use_fused is not None
-                    else _fused_step_available(p)
-                )
-                if (
-                    fused_ok
+                # Fused-path eligibility on the local shards (DTensor
+                # parameters under FSDP2 unwrap to their plain contiguous
+                # shard; the elementwise step on the shard IS the sharded
+                # step).
+                p_l = _local_view(p)
+                g_l = _local_view(grad)
+                m_l = _local_view(exp_avg)
+                v_l = _local_view(exp_avg_sq)
+                c_l = _local_view(state["compensation"]) if use_kahan else None
+                can_fuse = (
+                    self.use_fused is not False
                     and _fused_step_available(p)
-                    and p.is_contiguous()
-                    and exp_avg.is_contiguous()
-                    and exp_avg_sq.is_contiguous()
-                ):
+                    and p_l.is_contiguous()
+                    and g_l.is_contiguous()
+                    and m_l.is_contiguous()
+                    and v_l.is_contiguous()
+                    and (c_l is None or c_l.is_contiguous())
+                    and g_l.numel() == p_l.numel()
+                    and m_l.numel() == p_l.numel()
+                    and v_l.numel() == p_l.numel()
+                )
+                if self.use_fused is True and not can_fuse:
+                    raise RuntimeError(
+                        "use_fused=True, but the fused AnyPrecisionAdamW "
+                        "step cannot run for this parameter (kernel "
+                        "extension missing, non-GPU tensor, storage-less "
+                        "wrapper subclass, or non-contiguous/mismatched "
+                        "local shards). Pass use_fused=None to fall back "
+                        "to the eager op sequence automatically."
+                    )
+                if can_fuse:
                     from torchdistx_amd import _kernels
 
+                    self._fused_steps += 1
                     _kernels.anyprecision_adamw_(
-                        p,
-                        grad,
-                        exp_avg,
-                        exp_avg_sq,
-                        state["compensation"] if use_kahan else None,
+                        p_l,
+                        g_l,
+                        m_l,
+                        v_l,
+                        c_l,
                         lr,
                         beta1,
                         beta2,
