@@ -98,83 +98,150 @@ constexpr uint64_t kPhiloxStridePerOp = 4;
 // RNG session, one per outermost deferred_init on each thread (sessions
 // are per-thread by construction: deferred_level is thread_local).
 //
-//   * `pin_seed` is a 64-bit nonce hashed from the FULL default CPU
-//     generator state at session entry. Pins are a pure function of that
-//     state, so torch.manual_seed(S) fixes the native init bits, exactly
-//     like eager init is a function of the generator state at
-//     construction — and within one tape, replaying any subset on any
-//     rank gives identical bits (partition-invariant sharded
+// A session is partitioned into SEGMENTS by the mutations the user makes
+// to the default CPU generator. Recording never consumes the generator,
+// so at record time of each RNG op the full generator state is hashed:
+// an unchanged hash means "same segment, next slot", a changed hash
+// means the user re-seeded (or otherwise perturbed) the generator mid
+// construction and a new segment begins there. Each segment derives:
+//
+//   * a 64-bit Philox nonce = the state hash. Pins are therefore a pure
+//     function of the generator state in effect when the op was
+//     recorded: torch.manual_seed(S) — before deferred_init OR inside
+//     the module builder — fixes the native init bits, exactly like
+//     eager init is a function of the generator state at construction,
+//     and independently of any ambient (e.g. per-rank) seeding that the
+//     in-builder seed overrides. Within one tape, replaying any subset
+//     on any rank gives identical bits (partition-invariant sharded
 //     materialization).
-//   * Session entry then ADVANCES the live CPU generator by one 64-bit
-//     draw, the way eager construction consumes it: a second session
-//     without re-seeding hashes a different state and therefore draws an
-//     independent stream (two unseeded models never collide).
-//   * `replay_cursor` snapshots the pre-advance state. CPU stock replay
-//     of the session's RNG ops draws from this cursor (not the ambient
-//     generator), so a full in-tape-order CPU materialization is
-//     bitwise-equal to eager construction under the same seed, and replay
-//     never perturbs the ambient generator stream.
+//   * a state snapshot. CPU stock replay draws from a cursor initialized
+//     to the segment snapshot (not from the ambient generator), so a
+//     full in-tape-order CPU materialization is bitwise-equal to eager
+//     construction, and replay never perturbs the ambient stream.
+//
+// Session EXIT advances the live CPU generator by one 64-bit draw, the
+// way eager construction consumes it: a second session without
+// re-seeding hashes a different state and draws an independent stream
+// (two unseeded models never collide), while a session entered right
+// after torch.manual_seed(S) still sees the pristine state S — so its
+// pins and cursors match an eager construction from S bitwise.
 struct RngSession {
-  uint64_t pin_seed = 0;
-  uint64_t next_slot = 1;
-  // Default CPU generator state; advanced only by this session's CPU
-  // stock replay, in tape order. Guarded by tape_mutex (replay holds it).
+  bool has_segment = false;
+  uint64_t segment_hash = 0;   // raw state hash of the current segment
+  uint64_t nonce = 0;          // collision-salted Philox seed in use
+  uint64_t next_slot = 1;      // Philox slot within the current segment
+  std::vector<uint64_t> raw_hashes;  // for registry release at session end
+  // Default CPU generator state; seeded from segment snapshots and
+  // advanced only by this session's CPU stock replay, in tape order.
+  // Guarded by tape_mutex (replay holds it).
   at::Tensor replay_cursor;
 };
 
 thread_local std::shared_ptr<RngSession> current_rng_session;
 
+// Live-segment registry: two sessions OPEN AT THE SAME TIME (concurrent
+// threads) that hash the same generator state would otherwise pin the
+// same stream — recording consumes nothing, so the shared state cannot
+// tell them apart. Concurrent collisions get a salted nonce (and a
+// salted replay snapshot); sequential sessions release their entries at
+// exit, so re-seeding to the same state later still reproduces exactly.
+std::mutex segment_registry_mutex;
+std::unordered_map<uint64_t, uint64_t> live_segment_uses;
+
+uint64_t mixSalt(uint64_t h, uint64_t salt) {
+  for (int b = 0; b < 8; ++b) {
+    h = (h ^ ((salt >> (8 * b)) & 0xff)) * 1099511628211ull;
+  }
+  return h;
+}
+
 bool isRngOpName(const std::string& name) {
   return name == "aten::uniform_" || name == "aten::normal_";
 }
 
-void beginRngSession() {
-  auto session = std::make_shared<RngSession>();
-  // FNV-1a over the seed and the full generator state: equal state
-  // -> equal nonce; any consumption or re-seeding -> different nonce.
-  auto gen = at::globalContext().defaultGenerator(c10::DeviceType::CPU);
+// FNV-1a over a generator-state blob.
+uint64_t hashStateBytes(const uint8_t* bytes, int64_t n) {
   uint64_t h = 1469598103934665603ull;
-  auto mix = [&h](uint64_t v) {
-    for (int i = 0; i < 8; ++i) {
-      h = (h ^ ((v >> (8 * i)) & 0xff)) * 1099511628211ull;
+  uint64_t word = 0;
+  for (int64_t i = 0; i < n; ++i) {
+    word = (word << 8) | bytes[i];
+    if ((i & 7) == 7) {
+      for (int b = 0; b < 8; ++b) {
+        h = (h ^ ((word >> (8 * b)) & 0xff)) * 1099511628211ull;
+      }
+      word = 0;
     }
-  };
-  {
-    std::lock_guard<std::mutex> lock(gen.mutex());
-    mix(gen.current_seed());
-    at::Tensor state = gen.get_state();
-    const auto* bytes = state.const_data_ptr<uint8_t>();
-    const int64_t take = state.numel();
-    uint64_t word = 0;
-    for (int64_t i = 0; i < take; ++i) {
-      word = (word << 8) | bytes[i];
-      if ((i & 7) == 7) {
-        mix(word);
-        word = 0;
+  }
+  if ((n & 7) != 0) {
+    for (int b = 0; b < 8; ++b) {
+      h = (h ^ ((word >> (8 * b)) & 0xff)) * 1099511628211ull;
+    }
+  }
+  return h;
+}
+
+void beginRngSession() {
+  current_rng_session = std::make_shared<RngSession>();
+}
+
+void endRngSession() {
+  if (current_rng_session != nullptr) {
+    std::lock_guard<std::mutex> lock(segment_registry_mutex);
+    for (uint64_t h : current_rng_session->raw_hashes) {
+      auto it = live_segment_uses.find(h);
+      if (it != live_segment_uses.end() && --(it->second) == 0) {
+        live_segment_uses.erase(it);
       }
     }
-    if ((take & 7) != 0) {
-      mix(word);
-    }
-    session->replay_cursor = std::move(state);  // get_state returns a copy
-    // Consume one draw so the next unseeded session sees a new state.
-    gen.get<at::CPUGeneratorImpl>()->random64();
   }
-  session->pin_seed = h;
-  current_rng_session = std::move(session);
+  current_rng_session = nullptr;
+  // Consume one draw so the next unseeded session sees a new state
+  // (recorded ops keep the session alive through their own references).
+  auto gen = at::globalContext().defaultGenerator(c10::DeviceType::CPU);
+  std::lock_guard<std::mutex> lock(gen.mutex());
+  gen.get<at::CPUGeneratorImpl>()->random64();
 }
 
 // Pins (seed, counter-offset) for every recorded RNG op. The pins drive
 // the tdx Philox kernels on GPU replay and the slice-materialization fast
 // path on any device; plain CPU replay ignores them (it draws from the
-// session's replay cursor for eager bit-parity).
+// segment's replay cursor for eager bit-parity). When this op opens a new
+// segment, `segment_start` receives the segment's generator-state
+// snapshot (undefined otherwise).
 std::optional<std::pair<uint64_t, uint64_t>> pinPhiloxForOp(
-    const std::string& name) {
+    const std::string& name, at::Tensor* segment_start) {
   if (!isRngOpName(name) || current_rng_session == nullptr) {
     return std::nullopt;
   }
   RngSession& session = *current_rng_session;
-  return std::make_pair(session.pin_seed,
+  auto gen = at::globalContext().defaultGenerator(c10::DeviceType::CPU);
+  at::Tensor state;
+  {
+    std::lock_guard<std::mutex> lock(gen.mutex());
+    state = gen.get_state();  // returns a fresh copy
+  }
+  const uint64_t h =
+      hashStateBytes(state.const_data_ptr<uint8_t>(), state.numel());
+  if (!session.has_segment || h != session.segment_hash) {
+    session.has_segment = true;
+    session.segment_hash = h;
+    session.next_slot = 1;
+    uint64_t salt = 0;
+    {
+      std::lock_guard<std::mutex> lock(segment_registry_mutex);
+      salt = live_segment_uses[h]++;
+    }
+    session.raw_hashes.push_back(h);
+    session.nonce = salt == 0 ? h : mixSalt(h, salt);
+    if (salt != 0) {
+      // Concurrent collision: give the CPU stock path a distinct stream
+      // too, by snapshotting a generator seeded with the salted nonce.
+      auto salted = at::make_generator<at::CPUGeneratorImpl>(session.nonce);
+      state = salted.get_state();
+    }
+    *segment_start = std::move(state);
+  }
+  return std::make_pair(session.nonce,
                         session.next_slot++ * kPhiloxStridePerOp);
 }
 
@@ -448,13 +515,17 @@ void deferredInitHandler(const c10::OperatorHandle& op,
   });
 
   if (has_fake_arg || has_fake_ret) {
-    auto philox = pinPhiloxForOp(schema.operator_name().name);
-    // RNG ops additionally carry their session, for the CPU replay cursor.
+    at::Tensor segment_start;
+    auto philox = pinPhiloxForOp(schema.operator_name().name, &segment_start);
+    // RNG ops additionally carry their session (and, for the first op of
+    // a segment, the segment's generator-state snapshot) for the CPU
+    // replay cursor.
     std::shared_ptr<RngSession> session =
         philox.has_value() ? current_rng_session : nullptr;
     recordOp(
         schema.operator_name().name,
-        [handle = op, philox, session](torch::jit::Stack& s) {
+        [handle = op, philox, session,
+         segment_start](torch::jit::Stack& s) {
           // Replay hot path: recorded init ops whose target lives on the
           // GPU run through the hand-written CDNA4 kernels (tdx::) when
           // the _K extension is loaded. The pinned Philox state makes the
@@ -464,25 +535,33 @@ void deferredInitHandler(const c10::OperatorHandle& op,
           }
           if (session != nullptr && !s.empty() && s.front().isTensor()) {
             // CPU stock replay of a session RNG op: draw from the
-            // session's replay cursor instead of the ambient generator,
+            // segment's replay cursor instead of the ambient generator,
             // so in-tape-order materialization reproduces eager bits and
             // the ambient stream is left untouched. Ops recorded with an
             // explicit generator keep consuming that generator.
             const at::Tensor& self = s.front().toTensor();
             const bool explicit_gen = s.size() >= 4 && !s[3].isNone();
             if (self.defined() && self.is_cpu() && !explicit_gen) {
-              auto gen =
-                  at::globalContext().defaultGenerator(c10::DeviceType::CPU);
-              auto swap_state = [&gen](const at::Tensor& st) {
-                std::lock_guard<std::mutex> lock(gen.mutex());
-                at::Tensor prev = gen.get_state();
-                gen.set_state(st);
-                return prev;
-              };
-              at::Tensor ambient = swap_state(session->replay_cursor);
-              handle.callBoxed(s);
-              session->replay_cursor = swap_state(ambient);
-              return;
+              if (segment_start.defined()) {
+                session->replay_cursor = segment_start;
+              }
+              if (session->replay_cursor.defined()) {
+                auto gen = at::globalContext().defaultGenerator(
+                    c10::DeviceType::CPU);
+                auto swap_state = [&gen](const at::Tensor& st) {
+                  std::lock_guard<std::mutex> lock(gen.mutex());
+                  at::Tensor prev = gen.get_state();
+                  gen.set_state(st);
+                  return prev;
+                };
+                at::Tensor ambient = swap_state(session->replay_cursor);
+                handle.callBoxed(s);
+                session->replay_cursor = swap_state(ambient);
+                return;
+              }
+              // Out-of-tape-order replay reached a mid-segment op before
+              // its segment head: CPU bits for such partial replays are
+              // unspecified; draw from the ambient generator.
             }
           }
           handle.callBoxed(s);
@@ -861,6 +940,7 @@ void leaveDeferredInit() {
   if (--deferred_level == 0) {
     c10::impl::tls_set_dispatch_key_included(kDeferredKey, false);
     removeProxyHooks();
+    endRngSession();
   }
 }
 

@@ -174,32 +174,62 @@ struct FakeCall {
   std::unordered_map<c10::TensorImpl*, at::Tensor> meta_to_fake;
 };
 
+// A `device` schema argument names the output device only for ops that
+// actually consume one at dispatch time: a BackendSelect kernel, or a
+// TensorOptions argument pack (the consecutive dtype/layout/device/
+// pin_memory run codegen emits for factories). An argument that merely
+// happens to be named `device` is not a factory device (reference
+// fake.cc:370-414; documented rules 1-2 in the reference docs
+// fake_tensor_and_deferred_init.rst:120-137).
+bool opConsumesDeviceArg(const c10::OperatorHandle& op) {
+  if (op.hasKernelForDispatchKey(c10::DispatchKey::BackendSelect)) {
+    return true;
+  }
+  const auto& args = op.schema().arguments();
+  for (size_t i = 0; i + 3 < args.size(); ++i) {
+    if (args[i].name() == "dtype" && args[i + 1].name() == "layout" &&
+        args[i + 2].name() == "device" &&
+        args[i + 3].name() == "pin_memory") {
+      return true;
+    }
+  }
+  return false;
+}
+
 void assessOp(FakeCall& call) {
   const auto& schema = call.op.schema();
 
-  std::optional<c10::Device> fake_device;
-  visitTensors(*call.stack, call.args_begin, call.args_begin + call.num_args,
-               [&](const at::Tensor& t) {
-                 if (!t.defined()) {
-                   return;
-                 }
-                 call.has_tensor_arg = true;
-                 if (auto* fake = asFake(t)) {
-                   call.has_fake_arg = true;
-                   if (!fake_device.has_value()) {
-                     fake_device = fake->fake_device();
-                   }
-                 } else if (!call.tensor_device.has_value()) {
-                   call.tensor_device = t.device();
-                 }
-               });
-  // A fake argument's device wins over a real argument's device: an op
-  // mixing a fake "cuda" tensor with a real CPU scalar produces a fake
-  // "cuda" result (CPU-scalar exemption, reference fake.cc:314-316).
-  if (fake_device.has_value()) {
-    call.tensor_device = fake_device;
-  }
+  visitTensors(
+      *call.stack, call.args_begin, call.args_begin + call.num_args,
+      [&](const at::Tensor& t) {
+        if (!t.defined()) {
+          return;
+        }
+        call.has_tensor_arg = true;
+        if (isFake(t)) {
+          call.has_fake_arg = true;
+        }
+        // 0-dim CPU tensors are exempt from the common-device rule:
+        // they are ubiquitous as scalar operands of GPU expressions
+        // (reference fake.cc:314-316, 346-368).
+        if (t.dim() == 0 && t.device().is_cpu()) {
+          return;
+        }
+        c10::Device d = t.device();
+        if (!call.tensor_device.has_value()) {
+          call.tensor_device = d;
+        } else {
+          TORCH_CHECK(*call.tensor_device == d, "`", schema.operator_name(),
+                      "` was called on fake tensors with arguments spread "
+                      "over two devices (", *call.tensor_device, " and ", d,
+                      "); fake dispatch needs one common device (0-dim CPU "
+                      "scalars excepted).");
+        }
+      });
 
+  if (!opConsumesDeviceArg(call.op)) {
+    return;
+  }
   const auto& args = schema.arguments();
   for (size_t i = 0; i < args.size(); ++i) {
     if (args[i].name() == "device" && isDeviceType(args[i].type())) {
