@@ -489,3 +489,56 @@ def test_use_fused_true_raises_when_unavailable() -> None:
     opt = AnyPrecisionAdamW([p], lr=1e-3, use_fused=True)
     with pytest.raises(RuntimeError, match="use_fused=True"):
         opt.step()
+
+
+def test_parameter_subclass_is_preserved() -> None:
+    # Materialization must restore the ORIGINAL Python class, not degrade
+    # custom Parameter subclasses to plain nn.Parameter (mirrors the
+    # reference's tp_alloc class preservation).
+    class ScaledParameter(Parameter):
+        pass
+
+    class M(Module):
+        def __init__(self):
+            super().__init__()
+            self.p = ScaledParameter(torch.ones([4]))
+
+    m = deferred_init(M)
+    materialize_module(m)
+    assert type(m.p) is ScaledParameter
+    assert isinstance(m.p, Parameter)
+    assert m.p.requires_grad
+    assert torch.equal(m.p.detach(), torch.ones([4]))
+
+
+def test_tensor_subclass_buffer_is_preserved() -> None:
+    class TaggedTensor(Tensor):
+        pass
+
+    class M(Module):
+        def __init__(self):
+            super().__init__()
+            self.register_buffer("b", torch.zeros([3]).as_subclass(TaggedTensor))
+
+    m = deferred_init(M)
+    materialize_module(m)
+    assert type(m.b) is TaggedTensor
+    assert torch.equal(torch.Tensor(m.b), torch.zeros([3]))
+
+
+def test_aliased_subclass_parameters_materialize_to_one_object() -> None:
+    class ScaledParameter(Parameter):
+        pass
+
+    class M(Module):
+        def __init__(self):
+            super().__init__()
+            p = ScaledParameter(torch.ones([4]))
+            self.a = p
+            self.b = p
+
+    m = deferred_init(M)
+    a = materialize_tensor(cast(Tensor, m.a))
+    b = materialize_tensor(cast(Tensor, m.b))
+    assert a is b
+    assert type(a) is ScaledParameter
