@@ -41,6 +41,10 @@ def get_args():
                    choices=["replicate", "shard", "broadcast", "allgather", "slice"])
     p.add_argument("--dtype", type=str, default="bf16",
                    choices=["bf16", "fp32", "fp16"])
+    p.add_argument("--init", type=str, default="fast",
+                   choices=["fast", "stock"],
+                   help="fast: one normal_ per weight; stock: PyTorch's "
+                        "default kaiming resets (uniform_-dominated tape)")
     return p.parse_args()
 
 
@@ -86,6 +90,10 @@ def main():
 
     model_name = args.model or ("llama3-70b" if use_cuda else "tiny")
     cfg = CONFIGS[model_name]
+    if args.init != cfg.init:
+        import dataclasses
+
+        cfg = dataclasses.replace(cfg, init=args.init)
     dtype = {"bf16": torch.bfloat16, "fp32": torch.float32,
              "fp16": torch.float16}[args.dtype]
 
@@ -178,6 +186,7 @@ def main():
                 "model": cfg.name,
                 "n_params": n_params,
                 "mode": args.mode,
+                "init": args.init,
                 "device": "cuda" if use_cuda else "cpu",
                 "native_init_kernels": _kernels.available(),
                 "peak_host_rss_gb": round(rss_gb, 2),

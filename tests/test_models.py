@@ -70,3 +70,28 @@ def test_bf16_dtype_context() -> None:
     materialize_module(m)
     assert m.tok_emb.weight.dtype == torch.bfloat16
     assert torch.get_default_dtype() == torch.float32
+
+
+def test_stock_init_matches_eager_bitwise() -> None:
+    # cfg.init="stock" builds unmodified nn.Linear/nn.Embedding: the tape
+    # records PyTorch's kaiming resets (empty -> uniform_ chains) and CPU
+    # replay reproduces eager construction exactly.
+    import dataclasses
+
+    from torchdistx_amd import deferred_init
+    from torchdistx_amd.deferred_init import materialize_module
+    from torchdistx_amd.models import TINY, build_model
+
+    cfg = dataclasses.replace(TINY, init="stock")
+    torch.manual_seed(21)
+    m = deferred_init(build_model, cfg)
+    materialize_module(m)
+
+    torch.manual_seed(21)
+    e = build_model(cfg)
+    for (n1, p1), (n2, p2) in zip(m.named_parameters(), e.named_parameters()):
+        assert n1 == n2 and torch.equal(p1, p2), n1
+    # Stock init must actually produce plain nn.Linear modules.
+    from torchdistx_amd.models.transformer import InitLinear
+
+    assert not any(isinstance(mod, InitLinear) for mod in m.modules())

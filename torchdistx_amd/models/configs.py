@@ -24,6 +24,12 @@ class TransformerConfig:
     # MoE (Mixtral-style); None -> dense FFN
     moe_num_experts: Optional[int] = None
     moe_top_k: int = 2
+    # "fast": one normal_ per weight, zeros_ bias (llama-style, keeps each
+    # tape segment one RNG op long). "stock": PyTorch's default
+    # nn.Linear/nn.Embedding resets (kaiming_uniform_ weights + uniform_
+    # bias), which records the empty -> uniform_ tape shape of unmodified
+    # user models.
+    init: str = "fast"
 
     @property
     def n_params(self) -> int:

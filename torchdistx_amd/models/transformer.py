@@ -42,11 +42,28 @@ class InitEmbedding(nn.Embedding):
         self._fill_padding_idx_with_zero()
 
 
+import threading
+
+# Init mode in effect during model construction (set by build_model from
+# cfg.init). "stock" builds unmodified nn.Linear / nn.Embedding, whose
+# default resets record the empty -> kaiming(uniform_) tape shape of
+# ordinary user models; "fast" uses the single-normal_ variants above.
+_init_mode = threading.local()
+
+
+def _stock_init() -> bool:
+    return getattr(_init_mode, "mode", "fast") == "stock"
+
+
 def _linear(d_in: int, d_out: int, bias: bool = False) -> nn.Linear:
+    if _stock_init():
+        return nn.Linear(d_in, d_out, bias=bias)
     return InitLinear(d_in, d_out, bias=bias)
 
 
 def _embedding(num: int, dim: int) -> nn.Embedding:
+    if _stock_init():
+        return nn.Embedding(num, dim)
     return InitEmbedding(num, dim)
 
 
@@ -229,19 +246,24 @@ class TransformerLM(nn.Module):
 
 def build_model(cfg: TransformerConfig, device=None, dtype=None) -> TransformerLM:
     """Builds the model with the given default device/dtype; suitable for
-    use as the `module_fn` of deferred_init."""
-    if dtype is None and device is None:
-        return TransformerLM(cfg)
-    ctx_dtype = dtype if dtype is not None else torch.get_default_dtype()
-    prev_dtype = torch.get_default_dtype()
-    torch.set_default_dtype(ctx_dtype)
+    use as the `module_fn` of deferred_init. cfg.init selects between the
+    single-normal_ fast init and PyTorch's stock kaiming resets."""
+    _init_mode.mode = cfg.init
     try:
-        if device is not None:
-            with torch.device(device):
-                return TransformerLM(cfg)
-        return TransformerLM(cfg)
+        if dtype is None and device is None:
+            return TransformerLM(cfg)
+        ctx_dtype = dtype if dtype is not None else torch.get_default_dtype()
+        prev_dtype = torch.get_default_dtype()
+        torch.set_default_dtype(ctx_dtype)
+        try:
+            if device is not None:
+                with torch.device(device):
+                    return TransformerLM(cfg)
+            return TransformerLM(cfg)
+        finally:
+            torch.set_default_dtype(prev_dtype)
     finally:
-        torch.set_default_dtype(prev_dtype)
+        _init_mode.mode = "fast"
 
 
 __all__ = ["TransformerLM", "build_model", "RMSNorm", "MoELayer"]
