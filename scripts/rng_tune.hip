@@ -352,6 +352,76 @@ __global__ void normal_bf16_split2(__hip_bfloat16* __restrict__ out,
   }
 }
 
+// Combined candidate: rounds=7 Philox + log2 Box-Muller (the two wins
+// measured separately in round 1, never together).
+template <int R>
+__global__ void normal_bf16_rlog2(__hip_bfloat16* __restrict__ out,
+                                  uint32_t n8, float a, float b,
+                                  uint64_t seed, uint64_t offset) {
+  const uint32_t stride = gridDim.x * blockDim.x;
+  for (uint32_t g = blockIdx.x * blockDim.x + threadIdx.x; g < n8;
+       g += stride) {
+    uint4 bits = philoxR<R>(seed, g, offset);
+    uint32_t w[4] = {bits.x, bits.y, bits.z, bits.w};
+    V8 v;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float u1 = fmaxf((float)(w[j] & 0xffffu) * (1.0f / 65536.0f),
+                       1.1754944e-38f);
+      float r = sqrtf(-1.3862943611f * __log2f(u1));
+      float sn, cs;
+      __sincosf(
+          6.2831853071795865f * ((float)(w[j] >> 16) * (1.0f / 65536.0f)),
+          &sn, &cs);
+      v.v[j * 2 + 0] = __float2bfloat16(fmaf(r * cs, b, a));
+      v.v[j * 2 + 1] = __float2bfloat16(fmaf(r * sn, b, a));
+    }
+    *reinterpret_cast<V8*>(out + (uint64_t)g * 8) = v;
+  }
+}
+
+// r7 + log2 + split-phase x2 (transcendental chains of two groups
+// back-to-back for ILP).
+template <int R>
+__global__ void normal_bf16_rsplit2(__hip_bfloat16* __restrict__ out,
+                                    uint32_t n8, float a, float b,
+                                    uint64_t seed, uint64_t offset) {
+  const uint32_t stride = gridDim.x * blockDim.x;
+  const uint32_t half = (n8 + 1) / 2;
+  for (uint32_t t = blockIdx.x * blockDim.x + threadIdx.x; t < half;
+       t += stride) {
+    uint32_t g0 = t, g1 = t + half;
+    uint4 b0 = philoxR<R>(seed, g0, offset);
+    uint4 b1 = philoxR<R>(seed, g1, offset);
+    uint32_t w[8] = {b0.x, b0.y, b0.z, b0.w, b1.x, b1.y, b1.z, b1.w};
+    float r[8], sn[8], cs[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float u1 = fmaxf((float)(w[j] & 0xffffu) * (1.0f / 65536.0f),
+                       1.1754944e-38f);
+      r[j] = sqrtf(-1.3862943611f * __log2f(u1));
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      __sincosf(
+          6.2831853071795865f * ((float)(w[j] >> 16) * (1.0f / 65536.0f)),
+          &sn[j], &cs[j]);
+    }
+    V8 v0, v1;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      v0.v[j * 2 + 0] = __float2bfloat16(fmaf(r[j] * cs[j], b, a));
+      v0.v[j * 2 + 1] = __float2bfloat16(fmaf(r[j] * sn[j], b, a));
+      v1.v[j * 2 + 0] = __float2bfloat16(fmaf(r[4 + j] * cs[4 + j], b, a));
+      v1.v[j * 2 + 1] = __float2bfloat16(fmaf(r[4 + j] * sn[4 + j], b, a));
+    }
+    *reinterpret_cast<V8*>(out + (uint64_t)g0 * 8) = v0;
+    if (g1 < n8) {
+      *reinterpret_cast<V8*>(out + (uint64_t)g1 * 8) = v1;
+    }
+  }
+}
+
 template <typename K>
 double benchB(K kernel, __hip_bfloat16* buf, uint32_t n8, int blocks,
               int iters, int threads) {
@@ -414,6 +484,10 @@ int main() {
     printf("  normal  rounds=7     : %.2f TB/s\n", bench(normal_bf16_rounds<7>, buf, n8, blocks, 5));
     printf("  uniform rounds=10    : %.2f TB/s\n", bench(uniform_bf16_rounds<10>, buf, n8, blocks, 5));
     printf("  uniform rounds=7     : %.2f TB/s\n", bench(uniform_bf16_rounds<7>, buf, n8, blocks, 5));
+    printf("  normal  r7+log2      : %.2f TB/s\n", bench(normal_bf16_rlog2<7>, buf, n8, blocks, 5));
+    printf("  normal  r7+log2 t512 : %.2f TB/s\n", benchB(normal_bf16_rlog2<7>, buf, n8, blocks/2, 5, 512));
+    printf("  normal  r7+split2    : %.2f TB/s\n", bench(normal_bf16_rsplit2<7>, buf, n8, blocks, 5));
+    printf("  normal  r7+split2 512: %.2f TB/s\n", benchB(normal_bf16_rsplit2<7>, buf, n8, blocks/2, 5, 512));
   }
   // memset reference ceiling
   hipEvent_t e0, e1;

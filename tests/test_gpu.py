@@ -46,7 +46,9 @@ def test_tdx_uniform_statistics(dtype) -> None:
     assert f.var().item() == pytest.approx(64.0 / 12.0, rel=0.02)
 
 
-@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize(
+    "dtype", [torch.float32, torch.bfloat16, torch.float16]
+)
 def test_tdx_normal_statistics(dtype) -> None:
     n = 1 << 22
     t = torch.empty(n, device="cuda", dtype=dtype)
@@ -57,6 +59,28 @@ def test_tdx_normal_statistics(dtype) -> None:
     # ~0.27% of samples beyond 3 sigma
     frac3 = ((f - 1.5).abs() > 6.0).float().mean().item()
     assert 0.001 < frac3 < 0.006
+
+
+def test_tdx_normal_moments_and_ks() -> None:
+    # Distribution-quality gate for the tuned normal kernel: higher
+    # moments and a Kolmogorov-Smirnov test on a large fp32 sample.
+    from scipy import stats
+
+    n = 1 << 24
+    t = torch.empty(n, device="cuda", dtype=torch.float32)
+    torch.ops.tdx.normal_(t, 0.0, 1.0)
+    f = t.double()
+    m1 = f.mean().item()
+    m2 = f.var().item()
+    m3 = (f**3).mean().item()  # skewness (std normal: 0)
+    m4 = (f**4).mean().item()  # kurtosis (std normal: 3)
+    assert m1 == pytest.approx(0.0, abs=5e-3)
+    assert m2 == pytest.approx(1.0, rel=5e-3)
+    assert m3 == pytest.approx(0.0, abs=2e-2)
+    assert m4 == pytest.approx(3.0, rel=2e-2)
+    sample = t[:200000].cpu().numpy()
+    ks = stats.kstest(sample, "norm")
+    assert ks.pvalue > 1e-4, ks
 
 
 def test_tdx_uniform_seed_determinism() -> None:
