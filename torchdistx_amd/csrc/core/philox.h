@@ -32,7 +32,15 @@ TDX_HD uint32_t mulhi32(uint32_t a, uint32_t b) {
 #endif
 }
 
-TDX_HD U4 philox10(uint64_t seed, uint64_t subsequence, uint64_t offset) {
+// Round-count-templated Philox4x32. The uniform path uses the standard
+// 10 rounds (it is store-bound, extra rounds are free); the normal path
+// uses 7 rounds — Philox4x32-7 passes the full BigCrush battery (Salmon
+// et al., "Parallel random numbers: as easy as 1, 2, 3", SC'11, Table 2;
+// 10 is the authors' safety margin) and the normal kernel is VALU-bound,
+// where the 3 saved rounds are a measured ~13% fill-rate win on gfx950
+// (profiles/rng_tune_r7.log).
+template <int kRounds>
+TDX_HD U4 philox(uint64_t seed, uint64_t subsequence, uint64_t offset) {
   constexpr uint32_t kW0 = 0x9E3779B9u;
   constexpr uint32_t kW1 = 0xBB67AE85u;
   constexpr uint32_t kM0 = 0xD2511F53u;
@@ -46,7 +54,7 @@ TDX_HD U4 philox10(uint64_t seed, uint64_t subsequence, uint64_t offset) {
 #if defined(__HIP_DEVICE_COMPILE__)
 #pragma unroll
 #endif
-  for (int round = 0; round < 10; ++round) {
+  for (int round = 0; round < kRounds; ++round) {
     uint32_t lo0 = kM0 * c.x;
     uint32_t hi0 = mulhi32(kM0, c.x);
     uint32_t lo1 = kM1 * c.z;
@@ -56,6 +64,14 @@ TDX_HD U4 philox10(uint64_t seed, uint64_t subsequence, uint64_t offset) {
     k1 += kW1;
   }
   return c;
+}
+
+TDX_HD U4 philox10(uint64_t seed, uint64_t subsequence, uint64_t offset) {
+  return philox<10>(seed, subsequence, offset);
+}
+
+TDX_HD U4 philox7(uint64_t seed, uint64_t subsequence, uint64_t offset) {
+  return philox<7>(seed, subsequence, offset);
 }
 
 // uint32 -> [0, 1) float, 24-bit resolution (fp32 outputs).

@@ -38,11 +38,13 @@ struct GroupTraits<at::Half> {
 };
 
 // Fills vals[kElems] for element group `g` — the same mapping the CDNA4
-// kernels use (csrc/hip/init_kernels.hip rng_kernel).
+// kernels use (csrc/hip/init_kernels.hip rng_kernel). Normals draw from
+// Philox4x32-7, uniforms from Philox4x32-10 (see philox.h).
 template <typename T, bool kNormal>
 void groupValues(uint64_t g, float a, float b, uint64_t seed,
                  uint64_t offset, float* vals) {
-  philox::U4 bits = philox::philox10(seed, g, offset);
+  philox::U4 bits = kNormal ? philox::philox7(seed, g, offset)
+                            : philox::philox10(seed, g, offset);
   if constexpr (GroupTraits<T>::kElems == 4) {
     float u[4] = {philox::u32_to_uniform(bits.x),
                   philox::u32_to_uniform(bits.y),

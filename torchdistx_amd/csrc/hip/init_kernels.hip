@@ -2,7 +2,8 @@
 //
 // These are the hand-written HIP kernels behind the ops the deferred-init
 // tape records for standard module construction:
-// uniform_ / normal_ (Philox4x32-10 counter-based RNG), fill_ and zero_.
+// uniform_ / normal_ (counter-based Philox4x32: 10 rounds for uniforms,
+// 7 for the VALU-bound normals — see philoxN below), fill_ and zero_.
 // They register as the `tdx::` op namespace; the deferred-init replay
 // engine redirects the recorded aten:: init ops to them when the target
 // tensor lives on the GPU (csrc/core/native_redirect.cc).
@@ -56,9 +57,16 @@ __device__ __forceinline__ uint2 mulhilo32(uint32_t a, uint32_t b) {
   return r;
 }
 
-__device__ __forceinline__ uint4 philox10(uint64_t seed,
-                                          uint64_t subsequence,
-                                          uint64_t offset) {
+// Round-count-templated Philox4x32, kept in sync with the CPU reference
+// in csrc/core/philox.h. Uniforms use the standard 10 rounds (the kernel
+// is store-bound, extra rounds are free); normals use 7 — Philox4x32-7
+// passes the full BigCrush battery (Salmon et al., SC'11, Table 2) and
+// the transcendental-heavy normal kernel is VALU-bound, where the saved
+// rounds are a measured ~13% fill-rate win (profiles/rng_tune_r7.log).
+template <int kRounds>
+__device__ __forceinline__ uint4 philoxN(uint64_t seed,
+                                         uint64_t subsequence,
+                                         uint64_t offset) {
   constexpr uint32_t kW0 = 0x9E3779B9u;
   constexpr uint32_t kW1 = 0xBB67AE85u;
   constexpr uint32_t kM0 = 0xD2511F53u;
@@ -71,7 +79,7 @@ __device__ __forceinline__ uint4 philox10(uint64_t seed,
                        static_cast<uint32_t>(subsequence),
                        static_cast<uint32_t>(subsequence >> 32));
 #pragma unroll
-  for (int round = 0; round < 10; ++round) {
+  for (int round = 0; round < kRounds; ++round) {
     uint2 r0 = mulhilo32(kM0, c.x);
     uint2 r1 = mulhilo32(kM1, c.z);
     c = make_uint4(r1.y ^ c.y ^ k0, r1.x, r0.y ^ c.w ^ k1, r0.x);
@@ -157,7 +165,8 @@ __device__ __forceinline__ void rngGroupValues(uint64_t g,
                                                uint64_t seed,
                                                uint64_t offset,
                                                float* vals) {
-  uint4 bits = philox10(seed, g, offset);
+  uint4 bits = kDist == Dist::kNormal ? philoxN<7>(seed, g, offset)
+                                      : philoxN<10>(seed, g, offset);
   if constexpr (VecTraits<T>::kElems == 4) {
     float u[4] = {u32_to_uniform(bits.x), u32_to_uniform(bits.y),
                   u32_to_uniform(bits.z), u32_to_uniform(bits.w)};
