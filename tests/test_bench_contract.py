@@ -60,3 +60,33 @@ def test_bench_json_contract_gpu() -> None:
     assert d["config"]["device"] == "cuda"
     assert d["config"]["native_init_kernels"] is True
     assert d["value"] > 0
+
+
+def test_bench_selftest_world1() -> None:
+    result = subprocess.run(
+        [sys.executable, "bench.py", "--selftest"],
+        capture_output=True, text=True, timeout=300, cwd=REPO,
+    )
+    assert result.returncode == 0, result.stderr
+    d = json.loads(result.stdout.strip().splitlines()[-1])
+    assert d["selftest"] == "pass", d
+
+
+def test_bench_selftest_world2_torchrun() -> None:
+    # Rehearses the exact launcher + rank plumbing the driver uses for
+    # multi-GPU benches, over gloo on CPU: communicator creation, every
+    # materialization mode's cross-rank bitwise contract, shard coverage
+    # and slice reassembly at world 2.
+    result = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29779", "bench.py", "--selftest"],
+        capture_output=True, text=True, timeout=600, cwd=REPO,
+    )
+    assert result.returncode == 0, result.stderr[-2000:]
+    line = [l for l in result.stdout.splitlines() if '"selftest"' in l][-1]
+    d = json.loads(line)
+    assert d["selftest"] == "pass", d
+    names = {c["name"] for c in d["checks"]}
+    assert "broadcast-skewed-cross-rank" in names
+    assert "shard-coverage" in names
