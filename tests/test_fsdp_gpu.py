@@ -143,10 +143,10 @@ def test_fsdp2_fully_shard_deferred(nccl_world1) -> None:
 
 
 def test_fsdp2_anyprecision_optimizer_step(nccl_world1) -> None:
-    # Regression: FSDP2 DTensor parameters are storage-less wrapper
-    # subclasses; the fused HIP AdamW step must refuse them (null device
-    # pointer) and take the eager op sequence, which dispatches through
-    # DTensor correctly.
+    # FSDP2 DTensor parameters unwrap to their plain contiguous local
+    # shard for the fused HIP AdamW step — the elementwise update on the
+    # shard IS the sharded update — so the 3.4x single-kernel path must
+    # actually run (asserted via the _fused_steps counter).
     from torchdistx_amd import deferred_init
     from torchdistx_amd.models import TINY, build_model
     from torchdistx_amd.optimizers import AnyPrecisionAdamW
@@ -167,3 +167,6 @@ def test_fsdp2_anyprecision_optimizer_step(nccl_world1) -> None:
         optim.step()
     torch.cuda.synchronize()
     assert torch.isfinite(loss.detach()).item()
+    n_params = sum(1 for p in module.parameters() if p.grad is not None)
+    assert optim._fused_steps >= 3 * max(n_params - 1, 1), (
+        optim._fused_steps, n_params)
