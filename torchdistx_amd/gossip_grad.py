@@ -82,45 +82,48 @@ class GossipGraDState(default.DefaultState):
         proc_per_node=None,
         random_seed=2403,
     ):
-        if num_modules is None or num_modules < 1:
-            raise ValueError("`num_modules` should be a positive integer.")
+        if not num_modules or num_modules < 1:
+            raise ValueError(
+                f"num_modules must be >= 1 (got {num_modules!r}); pass "
+                "get_num_modules(fsdp_model) so the topology rotation "
+                "aligns with backward passes."
+            )
         self.num_modules = num_modules
         self.topology = topology or Topology.DISSEMINATION
 
-        if local_process_group is None and num_nodes is None:
+        if (local_process_group is None) != (num_nodes is None):
+            raise ValueError(
+                "pass local_process_group and num_nodes together, or "
+                "neither (then one subgroup per node is derived)."
+            )
+        if local_process_group is None:
             self.local_process_group, subgroups = dist.new_subgroups()
             self.num_nodes = len(subgroups)
-        elif local_process_group is None or num_nodes is None:
-            raise ValueError(
-                "`local_process_group` and `num_nodes` should be provided together."
-            )
         else:
             if num_nodes < 1:
-                raise ValueError("`num_nodes` should be equal to 1 or more.")
+                raise ValueError(f"num_nodes must be >= 1, got {num_nodes}")
             self.local_process_group = local_process_group
             self.num_nodes = num_nodes
 
-        if self.num_nodes % 2 != 0 and self.topology == Topology.CUBE:
+        if self.topology == Topology.CUBE and self.num_nodes % 2 != 0:
             raise ValueError(
-                "Current implementation doesn't support uneven number"
-                " of nodes for CUBE topology."
+                f"the CUBE (hypercube) topology needs an even node count; "
+                f"got {self.num_nodes}. Use DISSEMINATION for odd counts."
             )
 
         super().__init__(self.local_process_group)
 
-        self.proc_per_node = (
-            proc_per_node
-            if proc_per_node is not None
-            else self.local_process_group.size()
-        )
-        if self.proc_per_node < 1:
-            raise ValueError("`proc_per_node` should be equal to 1 or more.")
+        if proc_per_node is None:
+            proc_per_node = self.local_process_group.size()
+        if proc_per_node < 1:
+            raise ValueError(
+                f"proc_per_node must be >= 1, got {proc_per_node}"
+            )
+        self.proc_per_node = proc_per_node
 
-        self.master_process_group = (
-            master_process_group
-            if master_process_group is not None
-            else self._create_master_group()
-        )
+        if master_process_group is None:
+            master_process_group = self._create_master_group()
+        self.master_process_group = master_process_group
 
         self.random_seed = random_seed
         self.topologies = self._generate_topologies(self.random_seed)
@@ -163,7 +166,7 @@ def _get_send_recv_peers(state):
     DISSEMINATION: send to node_rank + 2**power, receive from
     node_rank - 2**power (mod num_nodes).
     """
-    assert state.gossip_period > 0, "`gossip_period` should be greater than 0."
+    assert state.gossip_period > 0, "gossip_period must be positive"
     power = (state.iter // state.num_modules) % state.gossip_period
     node_rank = state.cur_topology.index(state.rank)
 
@@ -188,31 +191,22 @@ def _gossip(state, grad, scaling_factor=0.5):
     if send_peer == INVALID_PEER or recv_peer == INVALID_PEER:
         return
 
-    assert send_peer != state.rank and recv_peer != state.rank, (
-        "Expected send and receive peers to differ from the current rank: "
-        f"(current rank is {state.rank}, `send_peer` is {send_peer} "
-        f"and `recv_peer` is {recv_peer})"
+    assert state.rank not in (send_peer, recv_peer), (
+        f"a rank must never gossip with itself: rank={state.rank}, "
+        f"send={send_peer}, recv={recv_peer}"
     )
-    assert isinstance(
-        state.master_process_group, ProcessGroup
-    ), "`master_process_group` is not an instance of `ProcessGroup`"
+    assert isinstance(state.master_process_group, ProcessGroup), (
+        "master_process_group must be a torch.distributed ProcessGroup"
+    )
 
+    group = state.master_process_group
     recv_grad = torch.empty_like(grad)
-    ops = [
-        dist.P2POp(
-            op=dist.isend,
-            tensor=grad,
-            peer=send_peer,
-            group=state.master_process_group,
-        ),
-        dist.P2POp(
-            op=dist.irecv,
-            tensor=recv_grad,
-            peer=recv_peer,
-            group=state.master_process_group,
-        ),
+    pair = [
+        dist.P2POp(op=dist.isend, tensor=grad, peer=send_peer, group=group),
+        dist.P2POp(op=dist.irecv, tensor=recv_grad, peer=recv_peer,
+                   group=group),
     ]
-    for req in dist.batch_isend_irecv(ops):
+    for req in dist.batch_isend_irecv(pair):
         req.wait()
     grad.add_(recv_grad).mul_(scaling_factor)
 

@@ -27,7 +27,9 @@ class SlowMoState(default.DefaultState):
     """
 
     def __init__(self, subgroup, sync_grads=True):
-        self.subgroup = subgroup if subgroup is not None else dist.new_subgroups()[0]
+        if subgroup is None:
+            subgroup, _ = dist.new_subgroups()
+        self.subgroup = subgroup
         super().__init__(self.subgroup)
         self.sync_grads = sync_grads
 
@@ -40,5 +42,6 @@ def slowmo_hook(state: SlowMoState, grad: torch.Tensor):
         state: hook configuration (subgroup + pre/post division factors).
         grad: the flat gradient of one FSDP unit for the local batch.
     """
-    if state.sync_grads:
-        default.allreduce_hook(state, grad)
+    if not state.sync_grads:
+        return
+    default.allreduce_hook(state, grad)

@@ -24,6 +24,18 @@ import torch
 import torch.distributed.algorithms.model_averaging.averagers as averagers
 
 
+def _check_groups_have_lr(optim: torch.optim.Optimizer) -> None:
+    """The slow-momentum update divides by each group's lr, so every group
+    must carry one (and the optimizer must have groups at all)."""
+    if not optim.param_groups:
+        raise ValueError("the base optimizer has no parameter groups")
+    if any("lr" not in g for g in optim.param_groups):
+        raise ValueError(
+            "every parameter group needs an explicit learning rate: the "
+            "slow-momentum outer update rescales by 1/lr per group"
+        )
+
+
 class SlowMomentumOptimizer(torch.optim.Optimizer):
     """Wraps a base optimizer and runs distributed training with Slow
     Momentum. Designed for FSDP modules with a ``NO_SHARD`` strategy together
@@ -45,34 +57,21 @@ class SlowMomentumOptimizer(torch.optim.Optimizer):
         slowmo_lr: float = 1.0,
     ):
         if base_optim is None:
-            raise ValueError("Base optimizer is a required parameter.")
+            raise ValueError("a base optimizer is required (got None)")
         self._base_optim = base_optim
-
-        if not self._base_optim.param_groups:
-            raise ValueError(
-                "Provided base optimizer does not have parameters specified."
-            )
-        for group in self._base_optim.param_groups:
-            if "lr" not in group:
-                raise ValueError(
-                    "All parameter groups should have learning rate specified."
-                )
+        _check_groups_have_lr(base_optim)
         self.param_groups = self._base_optim.param_groups
 
         if slowmo_freq < 1:
-            raise ValueError(
-                "Invalid ``slowmo_freq`` parameter, must be a positive value."
-            )
+            raise ValueError(f"slowmo_freq must be >= 1, got {slowmo_freq}")
         self.slowmo_freq = slowmo_freq
-
         if slowmo_factor < 0.0:
             raise ValueError(
-                "Invalid ``slowmo_factor`` parameter, must be non-negative."
+                f"slowmo_factor must be >= 0, got {slowmo_factor}"
             )
         self.slowmo_factor = slowmo_factor
-
         if slowmo_lr < 0.0:
-            raise ValueError("Invalid ``slowmo_lr`` parameter, must be non-negative.")
+            raise ValueError(f"slowmo_lr must be >= 0, got {slowmo_lr}")
         self.slowmo_lr = slowmo_lr
 
         self.averager = averagers.PeriodicModelAverager(
@@ -117,13 +116,7 @@ class SlowMomentumOptimizer(torch.optim.Optimizer):
         self.slowmo_lr = state_dict.pop("slowmo_lr")
         self.averager.step = state_dict.pop("step")
         self._base_optim.load_state_dict(state_dict)
-        if not self.param_groups:
-            raise ValueError("Base optimizer does not have parameter groups specified.")
-        for group in self._base_optim.param_groups:
-            if "lr" not in group:
-                raise ValueError(
-                    "All parameter groups should have learning rate specified."
-                )
+        _check_groups_have_lr(self._base_optim)
 
     @torch.no_grad()
     def step(self):
