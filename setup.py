@@ -18,6 +18,24 @@ os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 ROOT = os.path.dirname(os.path.abspath(__file__))
 CSRC = os.path.join(ROOT, "torchdistx_amd", "csrc")
 
+BASE_VERSION = "0.1.0"
+
+
+def version_with_variant() -> str:
+    """Version with a local variant tag naming the compute stack the wheel
+    was built against (rocmX.Y from the torch in the build env), so wheels
+    from different stacks are distinguishable — the same scheme the wheel
+    ecosystem uses for +cpu / +cuXXX / +rocmX.Y variants. Override with
+    TDX_VERSION_VARIANT (empty string = no tag, e.g. for sdists)."""
+    variant = os.environ.get("TDX_VERSION_VARIANT")
+    if variant is None:
+        hip = getattr(torch.version, "hip", None)
+        if hip:
+            variant = "rocm" + ".".join(hip.split(".")[:2])
+        else:
+            variant = "cpu"
+    return BASE_VERSION + ("+" + variant if variant else "")
+
 ext_modules = [
     cpp_extension.CppExtension(
         name="torchdistx_amd._C",
@@ -52,7 +70,7 @@ if all(os.path.exists(s) for s in hip_sources) and torch.version.hip:
 
 setup(
     name="torchdistx_amd",
-    version="0.1.0",
+    version=version_with_variant(),
     description=(
         "MI355X-native fake-tensor / deferred-init framework with "
         "torchdistx's capabilities"
