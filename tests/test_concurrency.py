@@ -91,3 +91,57 @@ def test_threaded_materialization_same_tensor() -> None:
 
     assert len(outs) == 16
     assert all(o is outs[0] for o in outs)
+
+
+def test_parallel_replay_speedup() -> None:
+    # The tape lock is released while a node's op executes, so disjoint
+    # parameters materialize genuinely in parallel. 8 threads over
+    # CPU-native Philox fills (no shared generator) must beat the
+    # single-thread wall clock clearly; >2x is expected on an 8-core box,
+    # 1.6x is the flake-proof gate.
+    import time
+
+    from torchdistx_amd import _C
+
+    def build():
+        # 32 independent 0.5M-element params: enough per-op work that
+        # execution dominates the (serialized) stack-building.
+        return nn.ParameterList(
+            [nn.Parameter(torch.empty(512, 1024).normal_()) for _ in range(32)]
+        )
+
+    def materialize_with(n_threads):
+        torch.manual_seed(123)
+        module = deferred_init(build)
+        params = list(module.parameters())
+        errors = []
+
+        def worker(chunk):
+            try:
+                for p in chunk:
+                    materialize_tensor(p)
+            except Exception as e:  # pragma: no cover
+                errors.append(e)
+
+        chunks = [params[i::n_threads] for i in range(n_threads)]
+        t0 = time.perf_counter()
+        threads = [
+            threading.Thread(target=worker, args=(c,)) for c in chunks
+        ]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
+        elapsed = time.perf_counter() - t0
+        assert not errors, errors
+        return elapsed
+
+    _C.set_native_init_cpu(True)
+    try:
+        materialize_with(1)  # warm caches/allocator
+        serial = min(materialize_with(1) for _ in range(3))
+        parallel = min(materialize_with(8) for _ in range(3))
+    finally:
+        _C.set_native_init_cpu(False)
+    speedup = serial / parallel
+    assert speedup > 1.6, f"parallel replay speedup only {speedup:.2f}x"

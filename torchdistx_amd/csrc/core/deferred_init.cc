@@ -20,6 +20,7 @@
 
 #include <algorithm>
 #include <atomic>
+#include <condition_variable>
 #include <functional>
 #include <memory>
 #include <mutex>
@@ -133,8 +134,13 @@ struct RngSession {
   std::vector<uint64_t> raw_hashes;  // for registry release at session end
   // Default CPU generator state; seeded from segment snapshots and
   // advanced only by this session's CPU stock replay, in tape order.
-  // Guarded by tape_mutex (replay holds it).
+  // Guarded by cursor_mutex: node execution runs without the tape lock,
+  // and the swap-draw-swap of the process-global default generator must
+  // be atomic per op (stock-CPU RNG replay therefore serializes per
+  // session — it consumes one generator stream by definition; the
+  // pinned-Philox native paths parallelize freely).
   at::Tensor replay_cursor;
+  std::mutex cursor_mutex;
 };
 
 thread_local std::shared_ptr<RngSession> current_rng_session;
@@ -275,17 +281,27 @@ struct OpNode {
   // writer of the same storage.
   std::vector<std::weak_ptr<AliasGroup>> write_groups;
   std::vector<at::Tensor> outputs;  // real tensors after replay
-  bool materialized = false;
+  bool materialized = false;  // Done: outputs valid, frame freed
+  // Another thread is executing this node's op with tape_mutex released;
+  // threads whose call stacks share the node wait on tape_cv instead of
+  // double-replaying. All transitions happen under tape_mutex.
+  bool running = false;
 };
 
 std::atomic<uint64_t> next_op_nr{1};
 
-// Serializes tape mutation (recordOp) and tape replay (materialize /
-// recordInfo): materialization runs with the GIL released, so without
-// this a second thread could record into or replay from the same node
-// graph mid-walk. Recursive because a terminal op (aten::item) can
-// materialize while a recording is on the stack.
+// Serializes tape STRUCTURE access: recording (recordOp), call-stack
+// construction, node state transitions and recordInfo. Materialization
+// runs with the GIL released, so without this a second thread could
+// record into or replay from the same node graph mid-walk. The lock is
+// NOT held while a node's op actually executes — replay releases it
+// around the kernel/compute work and marks the node `running`, so
+// disjoint subgraphs materialize concurrently on multiple threads (and
+// multiple HIP streams) while threads whose closures share a node wait
+// on tape_cv for its single execution. Recursive because a terminal op
+// (aten::item) can materialize while a recording is on the stack.
 std::recursive_mutex tape_mutex;
+std::condition_variable_any tape_cv;
 
 std::shared_ptr<TensorRecord> getRecord(FakeTensorImpl* fake) {
   return std::static_pointer_cast<TensorRecord>(fake->getData(kDeferredKey));
@@ -593,6 +609,7 @@ void deferredInitHandler(const c10::OperatorHandle& op,
             const at::Tensor& self = s.front().toTensor();
             const bool explicit_gen = s.size() >= 4 && !s[3].isNone();
             if (self.defined() && self.is_cpu() && !explicit_gen) {
+              std::lock_guard<std::mutex> cursor_lock{session->cursor_mutex};
               if (segment_start.defined()) {
                 session->replay_cursor = segment_start;
               }
@@ -949,22 +966,42 @@ torch::jit::Stack materializeArguments(OpNode& node) {
   return stack;
 }
 
-void replayNode(const std::shared_ptr<OpNode>& node) {
+// Ensures `node` is Done before returning. Called with `lock` held (at
+// recursion depth 1); releases the tape lock around the op execution so
+// other threads can build call stacks, record, and run disjoint nodes
+// concurrently. If another thread is already executing the node, waits
+// for it instead of double-replaying.
+void replayNode(std::unique_lock<std::recursive_mutex>& lock,
+                const std::shared_ptr<OpNode>& node) {
+  while (node->running) {
+    tape_cv.wait(lock);
+  }
   if (node->materialized) {
     return;
   }
   TORCH_CHECK(node->op.has_value(),
               "This deferred-init tape segment has already been freed and "
               "cannot be replayed again.");
+  // Dependencies are Done (the caller replays in chronological order and
+  // waits on in-flight nodes), so the argument stack can be built now.
   torch::jit::Stack stack = materializeArguments(*node);
-  {
+  node->running = true;
+  lock.unlock();
+  try {
     at::ThreadLocalStateGuard tls_guard{node->op->tls};
     // Replay must run for real regardless of any ambient fake/deferred
     // mode captured in the snapshot or active on this thread.
     c10::impl::ExcludeDispatchKeyGuard no_deferred{kDeferredKey};
     c10::impl::ExcludeDispatchKeyGuard no_fake{kFakeKey};
     node->op->run(stack);
+  } catch (...) {
+    lock.lock();
+    node->running = false;
+    tape_cv.notify_all();
+    throw;
   }
+  lock.lock();
+  node->running = false;
   node->outputs.clear();
   visitTensors(stack, 0, stack.size(),
                [&](const at::Tensor& t) { node->outputs.push_back(t); });
@@ -974,6 +1011,7 @@ void replayNode(const std::shared_ptr<OpNode>& node) {
   node->op.reset();
   node->input_slots.clear();
   node->input_slots.shrink_to_fit();
+  tape_cv.notify_all();
 }
 
 }  // namespace
@@ -1214,13 +1252,17 @@ at::Tensor materializeTensor(const at::Tensor& tensor) {
                     "`tensor` is fake but carries no deferred-init record, "
                     "so it cannot be materialized.");
 
-  std::lock_guard<std::recursive_mutex> lock{tape_mutex};
-  auto node = rec->desc.node;
+  std::unique_lock<std::recursive_mutex> lock{tape_mutex};
+  // Capture the producer descriptor while holding the lock: a concurrent
+  // recording thread may restamp `rec->desc` while this thread's replay
+  // runs unlocked, and the pre-restamp value is the one this call stack
+  // materializes (a legal serialization of the two racing calls).
+  OpOutputDescriptor desc = rec->desc;
   for (const auto& n : buildCallStack(*rec)) {
-    replayNode(n);
+    replayNode(lock, n);
   }
 
-  at::Tensor out = node->outputs.at(rec->desc.index);
+  at::Tensor out = desc.node->outputs.at(desc.index);
   // requires_grad_() is untraceable on purpose; re-apply the fake tensor's
   // final autograd flag explicitly (reference deferred_init.cc:713-729).
   if (out.is_leaf() && out.requires_grad() != tensor.requires_grad()) {
