@@ -45,6 +45,11 @@ def get_args():
                    choices=["fast", "stock"],
                    help="fast: one normal_ per weight; stock: PyTorch's "
                         "default kaiming resets (uniform_-dominated tape)")
+    p.add_argument("--threads", type=int, default=4,
+                   help="worker threads (one HIP stream each) for GPU "
+                        "replicate-mode materialization; the pinned Philox "
+                        "streams make the result bitwise independent of "
+                        "replay order. 1 = sequential.")
     p.add_argument("--selftest", action="store_true",
                    help="multi-rank preflight: validates communicator "
                         "creation, cross-rank bitwise equality of every "
@@ -133,6 +138,13 @@ def main():
             shards = materialize_module_dim0_sharded(module, rank, world)
             assert shards
             del shards
+        elif args.mode == "replicate" and use_cuda and args.threads > 1:
+            from torchdistx_amd.deferred_init import (
+                materialize_module_parallel,
+            )
+
+            materialize_module_parallel(module, num_threads=args.threads)
+            assert not is_deferred(module), "materialization incomplete"
         else:
             materialize_module_distributed(module, mode=args.mode)
             if args.mode != "shard":
@@ -197,6 +209,7 @@ def main():
                 "n_params": n_params,
                 "mode": args.mode,
                 "init": args.init,
+                "threads": args.threads,
                 "device": "cuda" if use_cuda else "cpu",
                 "native_init_kernels": _kernels.available(),
                 "peak_host_rss_gb": round(rss_gb, 2),

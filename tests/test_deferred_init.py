@@ -542,3 +542,17 @@ def test_aliased_subclass_parameters_materialize_to_one_object() -> None:
     b = materialize_tensor(cast(Tensor, m.b))
     assert a is b
     assert type(a) is ScaledParameter
+
+
+def test_materialize_module_parallel_cpu_fallback() -> None:
+    # On CPU targets the parallel API takes the sequential tape-order
+    # path, preserving the stock-generator eager-parity contract.
+    from torchdistx_amd.deferred_init import materialize_module_parallel
+
+    torch.manual_seed(77)
+    m = deferred_init(torch.nn.Linear, 8, 8)
+    materialize_module_parallel(m, num_threads=4)
+    torch.manual_seed(77)
+    e = torch.nn.Linear(8, 8)
+    assert torch.equal(m.weight, e.weight)
+    assert torch.equal(m.bias, e.bias)

@@ -323,3 +323,31 @@ def test_llama3_405b_rank_shard_fits_one_gpu() -> None:
     assert all(s.is_cuda for s in shards.values())
     del shards, m
     torch.cuda.empty_cache()
+
+
+def test_parallel_materialization_bitwise_equal() -> None:
+    # Thread-parallel materialization on per-thread HIP streams must be
+    # bitwise identical to the sequential path: the pinned Philox
+    # counters make the bits independent of replay order and stream.
+    from torchdistx_amd import deferred_init
+    from torchdistx_amd.deferred_init import (
+        materialize_module,
+        materialize_module_parallel,
+    )
+    from torchdistx_amd.models import TINY, build_model
+
+    torch.manual_seed(31)
+    seq = deferred_init(build_model, TINY, device="cuda", dtype=torch.bfloat16)
+    materialize_module(seq)
+
+    torch.manual_seed(31)
+    par = deferred_init(build_model, TINY, device="cuda", dtype=torch.bfloat16)
+    materialize_module_parallel(par, num_threads=4)
+    torch.cuda.synchronize()
+
+    for (n1, p1), (n2, p2) in zip(
+        seq.named_parameters(), par.named_parameters()
+    ):
+        assert n1 == n2 and torch.equal(p1, p2), n1
+    for (n1, b1), (n2, b2) in zip(seq.named_buffers(), par.named_buffers()):
+        assert n1 == n2 and torch.equal(b1, b2), n1
