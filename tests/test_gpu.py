@@ -351,3 +351,16 @@ def test_parallel_materialization_bitwise_equal() -> None:
         assert n1 == n2 and torch.equal(p1, p2), n1
     for (n1, b1), (n2, b2) in zip(seq.named_buffers(), par.named_buffers()):
         assert n1 == n2 and torch.equal(b1, b2), n1
+
+
+def test_tdx_bernoulli_statistics_and_shard() -> None:
+    n = 1 << 22
+    t = torch.empty(n, device="cuda", dtype=torch.float32)
+    torch.ops.tdx.bernoulli_(t, 0.25, seed=77, offset=4)
+    assert set(t.unique().tolist()) <= {0.0, 1.0}
+    assert t.mean().item() == pytest.approx(0.25, abs=0.002)
+    # Shard of the same stream is bitwise a slice of the full tensor.
+    shard = torch.empty(5000, device="cuda", dtype=torch.float32)
+    torch.ops.tdx.bernoulli_shard_(shard, 1234, 6234, 0.25, seed=77,
+                                   offset=4)
+    assert torch.equal(shard, t[1234:6234])

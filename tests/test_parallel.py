@@ -315,3 +315,89 @@ def test_slice_falls_back_for_explicit_generator() -> None:
     g2 = torch.Generator().manual_seed(3)
     expected = torch.empty(8, 4).uniform_(0, 1, generator=g2)
     assert torch.equal(shard.detach(), expected[:4])
+
+
+def test_slice_materialization_pointwise_chain() -> None:
+    # trunc_normal_-style init records uniform_ -> erfinv_ -> mul_ ->
+    # add_ -> clamp_; slicing commutes with the pointwise-scalar tail, so
+    # the shard path must reproduce the full native materialization
+    # bitwise on any row range.
+    from torch.nn import Module, Parameter
+
+    from torchdistx_amd import _C, deferred_init
+
+    class M(Module):
+        def __init__(self):
+            super().__init__()
+            w = torch.empty(64, 16)
+            torch.nn.init.trunc_normal_(w, mean=0.1, std=0.7, a=-1.0, b=1.5)
+            self.p = Parameter(w)
+
+    _C.set_native_init_cpu(True)
+    try:
+        torch.manual_seed(606)
+        full_m = deferred_init(M)
+        full = _C.materialize_tensor(full_m.p)
+
+        torch.manual_seed(606)
+        part_m = deferred_init(M)
+        for start, end in [(0, 64), (5, 17), (63, 64), (0, 1), (32, 32)]:
+            shard = _C.materialize_tensor_shard(part_m.p, start, end)
+            assert torch.equal(shard, full.detach()[start:end]), (start, end)
+    finally:
+        _C.set_native_init_cpu(False)
+
+
+def test_bernoulli_native_and_slice() -> None:
+    # bernoulli_ joins the pinned-Philox op set: native CPU replay draws
+    # from the pinned counters and the shard path reproduces any row
+    # range of a full materialization bitwise.
+    from torch.nn import Module, Parameter
+
+    from torchdistx_amd import _C, deferred_init
+
+    class M(Module):
+        def __init__(self):
+            super().__init__()
+            w = torch.empty(96, 32)
+            w.bernoulli_(0.3)
+            self.p = Parameter(w)
+
+    _C.set_native_init_cpu(True)
+    try:
+        torch.manual_seed(303)
+        full_m = deferred_init(M)
+        full = _C.materialize_tensor(full_m.p).detach()
+        vals = full.unique().tolist()
+        assert set(vals) <= {0.0, 1.0}
+        frac = full.mean().item()
+        assert 0.25 < frac < 0.35  # p = 0.3 over 3072 samples
+
+        torch.manual_seed(303)
+        part_m = deferred_init(M)
+        for start, end in [(0, 96), (7, 23), (95, 96)]:
+            shard = _C.materialize_tensor_shard(part_m.p, start, end)
+            assert torch.equal(shard, full[start:end]), (start, end)
+    finally:
+        _C.set_native_init_cpu(False)
+
+
+def test_bernoulli_stock_replay_matches_eager() -> None:
+    # Without the native path, bernoulli_ replays through the stock
+    # generator cursor and stays bitwise-equal to eager construction.
+    from torch.nn import Module, Parameter
+
+    from torchdistx_amd import deferred_init
+    from torchdistx_amd.deferred_init import materialize_module
+
+    class M(Module):
+        def __init__(self):
+            super().__init__()
+            self.p = Parameter(torch.empty(32, 8).bernoulli_(0.5))
+
+    torch.manual_seed(404)
+    m = deferred_init(M)
+    materialize_module(m)
+    torch.manual_seed(404)
+    e = M()
+    assert torch.equal(m.p, e.p)

@@ -162,7 +162,8 @@ uint64_t mixSalt(uint64_t h, uint64_t salt) {
 }
 
 bool isRngOpName(const std::string& name) {
-  return name == "aten::uniform_" || name == "aten::normal_";
+  return name == "aten::uniform_" || name == "aten::normal_" ||
+         name == "aten::bernoulli_";
 }
 
 // FNV-1a over a generator-state blob.
@@ -469,6 +470,9 @@ bool isInPlaceInitOp(const c10::OperatorName& name) {
   if (name.name == "aten::fill_") {
     return name.overload_name == "Scalar";
   }
+  if (name.name == "aten::bernoulli_") {
+    return name.overload_name == "float";
+  }
   return false;
 }
 
@@ -533,6 +537,12 @@ void deferredInitHandler(const c10::OperatorHandle& op,
         const double std = (*stack)[args_begin + 2].toDouble();
         TORCH_CHECK(std >= 0.0,
                     "normal_ expects std >= 0.0, but found std=", std);
+      }
+      if (opname == "aten::bernoulli_" && num_args >= 2 &&
+          (*stack)[args_begin + 1].isDouble()) {
+        const double p = (*stack)[args_begin + 1].toDouble();
+        TORCH_CHECK(0.0 <= p && p <= 1.0,
+                    "bernoulli_ expects 0 <= p <= 1, but found p=", p);
       }
       fast_inplace = true;
     }
@@ -607,7 +617,10 @@ void deferredInitHandler(const c10::OperatorHandle& op,
             // the ambient stream is left untouched. Ops recorded with an
             // explicit generator keep consuming that generator.
             const at::Tensor& self = s.front().toTensor();
-            const bool explicit_gen = s.size() >= 4 && !s[3].isNone();
+            bool explicit_gen = false;
+            for (const c10::IValue& v : s) {
+              explicit_gen = explicit_gen || v.isGenerator();
+            }
             if (self.defined() && self.is_cpu() && !explicit_gen) {
               std::lock_guard<std::mutex> cursor_lock{session->cursor_mutex};
               if (segment_start.defined()) {
@@ -1087,11 +1100,62 @@ bool chainNameIs(const std::string& name, const char* prefix) {
 
 // One value-affecting step of a simple init chain.
 struct ChainStep {
-  enum class Kind { kFactory, kUniform, kNormal, kFill, kZero, kPass } kind;
+  enum class Kind {
+    kFactory,
+    kUniform,
+    kNormal,
+    kBernoulli,
+    kFill,
+    kZero,
+    kPass,
+    // Elementwise op with only scalar parameters (erfinv_/mul_/add_/
+    // clamp_/...): slicing commutes with it, so a shard just applies the
+    // same op. This is what makes trunc_normal_-style init chains
+    // (uniform_ -> erfinv_ -> mul_ -> add_ -> clamp_) slice-
+    // materializable.
+    kPointwise,
+  } kind;
   double p0 = 0.0;  // from / mean / fill value
   double p1 = 0.0;  // to / std
   std::optional<std::pair<uint64_t, uint64_t>> philox;
+  std::vector<c10::IValue> pointwise_args;  // non-self args for kPointwise
+  // Replay closure of the recorded op (boxed call), for kPointwise.
+  std::function<void(torch::jit::Stack&)> pointwise_run;
 };
+
+// In-place elementwise ops whose only parameters are scalars; slicing
+// commutes with all of them.
+bool isPointwiseScalarOpName(const std::string& name) {
+  static const char* kNames[] = {
+      "aten::erfinv_",    "aten::mul_",       "aten::add_",
+      "aten::sub_",       "aten::div_",       "aten::clamp_",
+      "aten::clamp_min_", "aten::clamp_max_", "aten::neg_",
+      "aten::abs_",       "aten::sqrt_",      "aten::reciprocal_",
+  };
+  for (const char* n : kNames) {
+    if (name == n) {
+      return true;
+    }
+  }
+  return false;
+}
+
+bool isScalarLikeArg(const c10::IValue& v) {
+  if (v.isNone() || v.isScalar() || v.isDouble() || v.isInt() ||
+      v.isBool()) {
+    return true;
+  }
+  // Python scalars often reach the dispatcher as wrapped-number 0-dim
+  // tensors (`t.mul_(2.0)` -> mul_.Tensor). A real 0-dim tensor operand
+  // commutes with slicing just like a Scalar; an UNDEFINED tensor here
+  // is a nulled-out fake placeholder (a tensor dependency the shard path
+  // cannot substitute) and must be rejected.
+  if (v.isTensor()) {
+    const at::Tensor& t = v.toTensor();
+    return t.defined() && t.dim() == 0 && !isFake(t);
+  }
+  return false;
+}
 
 ChainStep classifyChainNode(const OpNode& node) {
   TORCH_CHECK(node.op.has_value(),
@@ -1125,16 +1189,24 @@ ChainStep classifyChainNode(const OpNode& node) {
   } else if (chainNameIs(op.name, "aten::full")) {
     step.kind = ChainStep::Kind::kFill;
     step.p0 = scalarArg(1);
-  } else if (op.name == "aten::uniform_" || op.name == "aten::normal_") {
-    TORCH_CHECK(op.args.size() < 4 || op.args[3].isNone(),
-                "slice materialization: `", op.name,
-                "` was recorded with an explicit generator, which the "
-                "counter-based shard path cannot honor; materialize the "
-                "tensor fully instead.");
-    step.kind = op.name == "aten::uniform_" ? ChainStep::Kind::kUniform
-                                            : ChainStep::Kind::kNormal;
-    step.p0 = scalarArg(1);
-    step.p1 = scalarArg(2);
+  } else if (op.name == "aten::uniform_" || op.name == "aten::normal_" ||
+             op.name == "aten::bernoulli_") {
+    for (size_t i = 1; i < op.args.size(); ++i) {
+      TORCH_CHECK(!op.args[i].isGenerator(),
+                  "slice materialization: `", op.name,
+                  "` was recorded with an explicit generator, which the "
+                  "counter-based shard path cannot honor; materialize the "
+                  "tensor fully instead.");
+    }
+    if (op.name == "aten::bernoulli_") {
+      step.kind = ChainStep::Kind::kBernoulli;
+      step.p0 = scalarArg(1);
+    } else {
+      step.kind = op.name == "aten::uniform_" ? ChainStep::Kind::kUniform
+                                              : ChainStep::Kind::kNormal;
+      step.p0 = scalarArg(1);
+      step.p1 = scalarArg(2);
+    }
     step.philox = op.philox;
   } else if (op.name == "aten::fill_") {
     step.kind = ChainStep::Kind::kFill;
@@ -1144,6 +1216,12 @@ ChainStep classifyChainNode(const OpNode& node) {
   } else if (op.name == "aten::detach" || op.name == "tdx::variable_data" ||
              op.name == "aten::alias") {
     step.kind = ChainStep::Kind::kPass;
+  } else if (isPointwiseScalarOpName(op.name) &&
+             std::all_of(op.args.begin() + 1, op.args.end(),
+                         isScalarLikeArg)) {
+    step.kind = ChainStep::Kind::kPointwise;
+    step.pointwise_args.assign(op.args.begin() + 1, op.args.end());
+    step.pointwise_run = op.run;
   } else {
     TORCH_CHECK(false, "slice materialization: `", op.name,
                 "` is not a whole-tensor init op; materialize the tensor "
@@ -1158,6 +1236,18 @@ void applyShardStep(const ChainStep& step, at::Tensor& shard, int64_t start,
     case ChainStep::Kind::kFactory:
     case ChainStep::Kind::kPass:
       return;
+    case ChainStep::Kind::kPointwise: {
+      torch::jit::Stack s;
+      s.emplace_back(shard);
+      for (const c10::IValue& a : step.pointwise_args) {
+        s.push_back(a);
+      }
+      c10::impl::ExcludeDispatchKeyGuard no_deferred{kDeferredKey};
+      c10::impl::ExcludeDispatchKeyGuard no_fake{kFakeKey};
+      step.pointwise_run(s);
+      shard = s.back().toTensor();
+      return;
+    }
     case ChainStep::Kind::kZero:
       shard.zero_();
       return;
@@ -1165,7 +1255,8 @@ void applyShardStep(const ChainStep& step, at::Tensor& shard, int64_t start,
       shard.fill_(step.p0);
       return;
     case ChainStep::Kind::kUniform:
-    case ChainStep::Kind::kNormal: {
+    case ChainStep::Kind::kNormal:
+    case ChainStep::Kind::kBernoulli: {
       TORCH_CHECK(step.philox.has_value(),
                   "slice materialization: this RNG op carries no pinned "
                   "Philox state (was it recorded by an older build?)");
@@ -1179,15 +1270,24 @@ void applyShardStep(const ChainStep& step, at::Tensor& shard, int64_t start,
               .findSchemaOrThrow("tdx::normal_shard_", "")
               .typed<at::Tensor&(at::Tensor&, int64_t, int64_t, double,
                                  double, int64_t, int64_t)>();
+      static const auto bernoulli_shard =
+          c10::Dispatcher::singleton()
+              .findSchemaOrThrow("tdx::bernoulli_shard_", "")
+              .typed<at::Tensor&(at::Tensor&, int64_t, int64_t, double,
+                                 int64_t, int64_t)>();
       const auto& [seed, offset] = *step.philox;
       if (step.kind == ChainStep::Kind::kUniform) {
         uniform_shard.call(shard, start, end, step.p0, step.p1,
                            static_cast<int64_t>(seed),
                            static_cast<int64_t>(offset));
-      } else {
+      } else if (step.kind == ChainStep::Kind::kNormal) {
         normal_shard.call(shard, start, end, step.p0, step.p1,
                           static_cast<int64_t>(seed),
                           static_cast<int64_t>(offset));
+      } else {
+        bernoulli_shard.call(shard, start, end, step.p0,
+                             static_cast<int64_t>(seed),
+                             static_cast<int64_t>(offset));
       }
       return;
     }

@@ -20,6 +20,7 @@ struct Redirect {
 constexpr Redirect kRedirects[] = {
     {"aten::uniform_", "", "uniform_", true},
     {"aten::normal_", "", "normal_", true},
+    {"aten::bernoulli_", "float", "bernoulli_", true},
     {"aten::fill_", "Scalar", "fill_", true},
     {"aten::zero_", "", "zero_", false},
     {"aten::copy_", "", "copy_", true},
@@ -110,12 +111,17 @@ bool tryNativeInitRedirect(
     return false;
   }
   const bool is_rng = std::strcmp(redirect->tdx_name, "uniform_") == 0 ||
-                      std::strcmp(redirect->tdx_name, "normal_") == 0;
+                      std::strcmp(redirect->tdx_name, "normal_") == 0 ||
+                      std::strcmp(redirect->tdx_name, "bernoulli_") == 0;
   if (is_rng) {
     // An explicitly passed generator must be honored; only the default
     // generator's stream is replaced by the pinned counter-based one.
-    if (stack.size() >= 4 && !stack[3].isNone()) {
-      return false;
+    // The generator slot differs per op (uniform_/normal_: arg 3,
+    // bernoulli_.float: arg 2), so scan for one.
+    for (const c10::IValue& v : stack) {
+      if (v.isGenerator()) {
+        return false;
+      }
     }
     // tdx RNG schemas carry two trailing optional args: the Philox seed
     // and counter offset pinned at record time (partition-invariant init).

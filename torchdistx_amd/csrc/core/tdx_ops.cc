@@ -37,12 +37,16 @@ struct GroupTraits<at::Half> {
   static constexpr int kElems = 8;
 };
 
+// RNG kinds shared with the CDNA4 kernels.
+enum class RngKind { kUniform, kNormal, kBernoulli };
+
 // Fills vals[kElems] for element group `g` — the same mapping the CDNA4
 // kernels use (csrc/hip/init_kernels.hip rng_kernel). Normals draw from
-// Philox4x32-7, uniforms from Philox4x32-10 (see philox.h).
-template <typename T, bool kNormal>
+// Philox4x32-7, uniforms and bernoulli from Philox4x32-10 (see philox.h).
+template <typename T, RngKind kKind>
 void groupValues(uint64_t g, float a, float b, uint64_t seed,
                  uint64_t offset, float* vals) {
+  constexpr bool kNormal = kKind == RngKind::kNormal;
   philox::U4 bits = kNormal ? philox::philox7(seed, g, offset)
                             : philox::philox10(seed, g, offset);
   if constexpr (GroupTraits<T>::kElems == 4) {
@@ -50,9 +54,13 @@ void groupValues(uint64_t g, float a, float b, uint64_t seed,
                   philox::u32_to_uniform(bits.y),
                   philox::u32_to_uniform(bits.z),
                   philox::u32_to_uniform(bits.w)};
-    if constexpr (!kNormal) {
+    if constexpr (kKind == RngKind::kUniform) {
       for (int j = 0; j < 4; ++j) {
         vals[j] = std::fmaf(u[j], b, a);
+      }
+    } else if constexpr (kKind == RngKind::kBernoulli) {
+      for (int j = 0; j < 4; ++j) {
+        vals[j] = u[j] < a ? 1.0f : 0.0f;
       }
     } else {
       for (int p = 0; p < 2; ++p) {
@@ -68,9 +76,12 @@ void groupValues(uint64_t g, float a, float b, uint64_t seed,
     for (int j = 0; j < 4; ++j) {
       float lo = philox::u16_to_uniform(words[j]);
       float hi = philox::u16_to_uniform(words[j] >> 16);
-      if constexpr (!kNormal) {
+      if constexpr (kKind == RngKind::kUniform) {
         vals[j * 2 + 0] = std::fmaf(lo, b, a);
         vals[j * 2 + 1] = std::fmaf(hi, b, a);
+      } else if constexpr (kKind == RngKind::kBernoulli) {
+        vals[j * 2 + 0] = lo < a ? 1.0f : 0.0f;
+        vals[j * 2 + 1] = hi < a ? 1.0f : 0.0f;
       } else {
         float u1 = std::max(lo, 1.1754944e-38f);
         float r = std::sqrt(-2.0f * std::log(u1));
@@ -84,7 +95,7 @@ void groupValues(uint64_t g, float a, float b, uint64_t seed,
 
 // Writes elements [start, end) of the virtual full tensor into
 // out[0 .. end-start), reproducing the group-indexed counter layout.
-template <typename T, bool kNormal>
+template <typename T, RngKind kKind>
 void cpuPhiloxRange(T* out, int64_t start, int64_t end, float a, float b,
                     uint64_t seed, uint64_t offset) {
   constexpr int kElems = GroupTraits<T>::kElems;
@@ -92,8 +103,8 @@ void cpuPhiloxRange(T* out, int64_t start, int64_t end, float a, float b,
   const int64_t g_last = (end + kElems - 1) / kElems;
   float vals[kElems];
   for (int64_t g = g_first; g < g_last; ++g) {
-    groupValues<T, kNormal>(static_cast<uint64_t>(g), a, b, seed, offset,
-                            vals);
+    groupValues<T, kKind>(static_cast<uint64_t>(g), a, b, seed, offset,
+                          vals);
     const int64_t base = g * kElems;
     const int64_t lo = std::max(base, start);
     const int64_t hi = std::min(base + kElems, end);
@@ -103,7 +114,7 @@ void cpuPhiloxRange(T* out, int64_t start, int64_t end, float a, float b,
   }
 }
 
-template <bool kNormal>
+template <RngKind kKind>
 void cpuPhiloxDispatch(at::Tensor& self, int64_t start, int64_t end,
                        float a, float b, uint64_t seed, uint64_t offset) {
   TORCH_CHECK(self.is_contiguous(),
@@ -112,15 +123,15 @@ void cpuPhiloxDispatch(at::Tensor& self, int64_t start, int64_t end,
               "shard numel must equal end - start");
   switch (self.scalar_type()) {
     case at::kFloat:
-      cpuPhiloxRange<float, kNormal>(self.data_ptr<float>(), start, end, a,
+      cpuPhiloxRange<float, kKind>(self.data_ptr<float>(), start, end, a,
                                      b, seed, offset);
       break;
     case at::kBFloat16:
-      cpuPhiloxRange<at::BFloat16, kNormal>(self.data_ptr<at::BFloat16>(),
+      cpuPhiloxRange<at::BFloat16, kKind>(self.data_ptr<at::BFloat16>(),
                                             start, end, a, b, seed, offset);
       break;
     case at::kHalf:
-      cpuPhiloxRange<at::Half, kNormal>(self.data_ptr<at::Half>(), start,
+      cpuPhiloxRange<at::Half, kKind>(self.data_ptr<at::Half>(), start,
                                         end, a, b, seed, offset);
       break;
     default:
@@ -136,7 +147,7 @@ at::Tensor& cpu_uniform_(at::Tensor& self, double from, double to,
                          std::optional<int64_t> seed,
                          std::optional<int64_t> offset) {
   if (seed.has_value() && offset.has_value()) {
-    cpuPhiloxDispatch<false>(self, 0, self.numel(),
+    cpuPhiloxDispatch<RngKind::kUniform>(self, 0, self.numel(),
                              static_cast<float>(from),
                              static_cast<float>(to - from),
                              static_cast<uint64_t>(*seed),
@@ -152,13 +163,28 @@ at::Tensor& cpu_normal_(at::Tensor& self, double mean, double std,
                         std::optional<int64_t> offset) {
   TORCH_CHECK(std >= 0.0, "normal_ expects std >= 0.0, but found std=", std);
   if (seed.has_value() && offset.has_value()) {
-    cpuPhiloxDispatch<true>(self, 0, self.numel(), static_cast<float>(mean),
+    cpuPhiloxDispatch<RngKind::kNormal>(self, 0, self.numel(), static_cast<float>(mean),
                             static_cast<float>(std),
                             static_cast<uint64_t>(*seed),
                             static_cast<uint64_t>(*offset));
     return self;
   }
   return self.normal_(mean, std, std::move(generator));
+}
+
+at::Tensor& cpu_bernoulli_(at::Tensor& self, double p,
+                           std::optional<at::Generator> generator,
+                           std::optional<int64_t> seed,
+                           std::optional<int64_t> offset) {
+  TORCH_CHECK(0.0 <= p && p <= 1.0,
+              "bernoulli_ expects 0 <= p <= 1, but found p=", p);
+  if (seed.has_value() && offset.has_value()) {
+    cpuPhiloxDispatch<RngKind::kBernoulli>(
+        self, 0, self.numel(), static_cast<float>(p), 0.0f,
+        static_cast<uint64_t>(*seed), static_cast<uint64_t>(*offset));
+    return self;
+  }
+  return self.bernoulli_(p, std::move(generator));
 }
 
 at::Tensor& cpu_fill_(at::Tensor& self, const at::Scalar& value) {
@@ -179,7 +205,7 @@ at::Tensor& cpu_copy_(at::Tensor& self, const at::Tensor& src,
 at::Tensor& cpu_uniform_shard_(at::Tensor& shard, int64_t start, int64_t end,
                                double from, double to, int64_t seed,
                                int64_t offset) {
-  cpuPhiloxDispatch<false>(shard, start, end, static_cast<float>(from),
+  cpuPhiloxDispatch<RngKind::kUniform>(shard, start, end, static_cast<float>(from),
                            static_cast<float>(to - from),
                            static_cast<uint64_t>(seed),
                            static_cast<uint64_t>(offset));
@@ -189,10 +215,19 @@ at::Tensor& cpu_uniform_shard_(at::Tensor& shard, int64_t start, int64_t end,
 at::Tensor& cpu_normal_shard_(at::Tensor& shard, int64_t start, int64_t end,
                               double mean, double std, int64_t seed,
                               int64_t offset) {
-  cpuPhiloxDispatch<true>(shard, start, end, static_cast<float>(mean),
+  cpuPhiloxDispatch<RngKind::kNormal>(shard, start, end, static_cast<float>(mean),
                           static_cast<float>(std),
                           static_cast<uint64_t>(seed),
                           static_cast<uint64_t>(offset));
+  return shard;
+}
+
+at::Tensor& cpu_bernoulli_shard_(at::Tensor& shard, int64_t start,
+                                 int64_t end, double p, int64_t seed,
+                                 int64_t offset) {
+  cpuPhiloxDispatch<RngKind::kBernoulli>(
+      shard, start, end, static_cast<float>(p), 0.0f,
+      static_cast<uint64_t>(seed), static_cast<uint64_t>(offset));
   return shard;
 }
 
@@ -203,6 +238,10 @@ TORCH_LIBRARY(tdx, m) {
       "-> Tensor(a!)");
   m.def(
       "normal_(Tensor(a!) self, float mean=0., float std=1., *, "
+      "Generator? generator=None, int? seed=None, int? offset=None) "
+      "-> Tensor(a!)");
+  m.def(
+      "bernoulli_(Tensor(a!) self, float p=0.5, *, "
       "Generator? generator=None, int? seed=None, int? offset=None) "
       "-> Tensor(a!)");
   m.def("fill_(Tensor(a!) self, Scalar value) -> Tensor(a!)");
@@ -216,6 +255,9 @@ TORCH_LIBRARY(tdx, m) {
   m.def(
       "normal_shard_(Tensor(a!) shard, int start, int end, float mean=0., "
       "float std=1., *, int seed, int offset) -> Tensor(a!)");
+  m.def(
+      "bernoulli_shard_(Tensor(a!) shard, int start, int end, float p=0.5, "
+      "*, int seed, int offset) -> Tensor(a!)");
 }
 
 TORCH_LIBRARY_IMPL(tdx, CPU, m) {
@@ -226,6 +268,8 @@ TORCH_LIBRARY_IMPL(tdx, CPU, m) {
   m.impl("copy_", cpu_copy_);
   m.impl("uniform_shard_", cpu_uniform_shard_);
   m.impl("normal_shard_", cpu_normal_shard_);
+  m.impl("bernoulli_", cpu_bernoulli_);
+  m.impl("bernoulli_shard_", cpu_bernoulli_shard_);
 }
 
 }  // namespace

@@ -151,7 +151,7 @@ struct VecTraits<__half> {
   };
 };
 
-enum class Dist { kUniform, kNormal };
+enum class Dist { kUniform, kNormal, kBernoulli };
 
 // Computes the VecTraits<T>::kElems values of element group `g` — ONE
 // philox10 with counter = the 16-byte group index turned into 4 fp32
@@ -175,6 +175,11 @@ __device__ __forceinline__ void rngGroupValues(uint64_t g,
       for (int j = 0; j < 4; ++j) {
         vals[j] = fmaf(u[j], b, a);
       }
+    } else if constexpr (kDist == Dist::kBernoulli) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        vals[j] = u[j] < a ? 1.0f : 0.0f;
+      }
     } else {
       float2 n01 = box_muller(u[0], u[1]);
       float2 n23 = box_muller(u[2], u[3]);
@@ -190,6 +195,12 @@ __device__ __forceinline__ void rngGroupValues(uint64_t g,
       for (int j = 0; j < 4; ++j) {
         vals[j * 2 + 0] = fmaf(u16_to_uniform(words[j]), b, a);
         vals[j * 2 + 1] = fmaf(u16_to_uniform(words[j] >> 16), b, a);
+      }
+    } else if constexpr (kDist == Dist::kBernoulli) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        vals[j * 2 + 0] = u16_to_uniform(words[j]) < a ? 1.0f : 0.0f;
+        vals[j * 2 + 1] = u16_to_uniform(words[j] >> 16) < a ? 1.0f : 0.0f;
       }
     } else {
 #pragma unroll
@@ -310,8 +321,9 @@ void launchRng(at::Tensor& self,
   }
   auto stream = at::cuda::getCurrentCUDAStream();
 
-  // p0/p1 are (from, to) for uniform and (mean, std) for normal; the kernel
-  // applies a + b * sample with b = range or std respectively.
+  // p0/p1 are (from, to) for uniform, (mean, std) for normal, and
+  // (p, unused) for bernoulli; uniform/normal apply a + b * sample,
+  // bernoulli thresholds the raw uniform against a.
   float a = static_cast<float>(p0);
   float b = kDist == Dist::kUniform ? static_cast<float>(p1 - p0)
                                     : static_cast<float>(p1);
@@ -481,6 +493,26 @@ at::Tensor& tdx_normal_shard_(at::Tensor& shard, int64_t start, int64_t end,
                               int64_t offset) {
   TORCH_CHECK(std >= 0.0, "normal_ expects std >= 0.0, but found std=", std);
   launchRngShard<Dist::kNormal>(shard, start, end, mean, std, seed, offset);
+  return shard;
+}
+
+at::Tensor& tdx_bernoulli_(at::Tensor& self,
+                           double p,
+                           std::optional<at::Generator> generator,
+                           std::optional<int64_t> seed,
+                           std::optional<int64_t> offset) {
+  TORCH_CHECK(0.0 <= p && p <= 1.0,
+              "bernoulli_ expects 0 <= p <= 1, but found p=", p);
+  launchRng<Dist::kBernoulli>(self, p, 0.0, generator, seed, offset);
+  return self;
+}
+
+at::Tensor& tdx_bernoulli_shard_(at::Tensor& shard, int64_t start,
+                                 int64_t end, double p, int64_t seed,
+                                 int64_t offset) {
+  TORCH_CHECK(0.0 <= p && p <= 1.0,
+              "bernoulli_ expects 0 <= p <= 1, but found p=", p);
+  launchRngShard<Dist::kBernoulli>(shard, start, end, p, 0.0, seed, offset);
   return shard;
 }
 
@@ -656,6 +688,8 @@ TORCH_LIBRARY_IMPL(tdx, CUDA, m) {
   m.impl("copy_", tdx_copy_);
   m.impl("uniform_shard_", tdx_uniform_shard_);
   m.impl("normal_shard_", tdx_normal_shard_);
+  m.impl("bernoulli_", tdx_bernoulli_);
+  m.impl("bernoulli_shard_", tdx_bernoulli_shard_);
 }
 
 }  // namespace
