@@ -36,3 +36,40 @@ def bench(use_fused, steps=10):
 for name, fused in (("eager", False), ("fused", True)):
     ms, tb = bench(fused)
     print(f"{name}: {ms:8.2f} ms/step  effective {tb:5.2f} TB/s")
+
+
+def bench_many_small(use_batched, n_tensors=512, numel=16384, steps=10):
+    """Launch-overhead regime: hundreds of small params per step."""
+    torch.manual_seed(1)
+    params = [
+        torch.nn.Parameter(
+            torch.randn(numel, device="cuda", dtype=torch.bfloat16)
+        )
+        for _ in range(n_tensors)
+    ]
+    import torchdistx_amd.optimizers.anyprecision_optimizer as apo
+
+    saved = apo._BATCH_MAX_NUMEL
+    apo._BATCH_MAX_NUMEL = (1 << 20) if use_batched else 0
+    try:
+        opt = AnyPrecisionAdamW(
+            params, lr=1e-3, weight_decay=0.01, use_kahan_summation=True,
+            momentum_dtype=torch.float32, variance_dtype=torch.bfloat16,
+            compensation_buffer_dtype=torch.bfloat16,
+        )
+        for p in params:
+            p.grad = torch.randn_like(p)
+        opt.step()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(steps):
+            opt.step()
+        torch.cuda.synchronize()
+        return (time.perf_counter() - t0) / steps * 1e3
+    finally:
+        apo._BATCH_MAX_NUMEL = saved
+
+
+for name, batched in (("per-tensor fused", False), ("batched", True)):
+    ms = bench_many_small(batched)
+    print(f"512x16k params, {name}: {ms:8.2f} ms/step")

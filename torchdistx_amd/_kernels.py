@@ -56,3 +56,35 @@ def anyprecision_adamw_(
         step_size,
         bias_correction2_sqrt,
     )
+
+
+def anyprecision_adamw_batched_(
+    params,
+    grads,
+    exp_avgs,
+    exp_avg_sqs,
+    compensations,
+    lr,
+    beta1,
+    beta2,
+    eps,
+    weight_decay,
+    step_sizes,
+    bias_correction2_sqrts,
+):
+    """One kernel launch updating every listed parameter (see
+    csrc/hip/anyprecision_adamw.hip, adamw_batched_kernel)."""
+    _K.anyprecision_adamw_batched_(
+        params,
+        [g.contiguous() for g in grads],
+        exp_avgs,
+        exp_avg_sqs,
+        compensations,
+        lr,
+        beta1,
+        beta2,
+        eps,
+        weight_decay,
+        step_sizes,
+        bias_correction2_sqrts,
+    )

@@ -306,6 +306,95 @@ int64_t tryLaunchVec(const AdamWArgs& args, int grid, int block,
   return launched ? n_full : 0;
 }
 
+// ---------------------------------------------------------------------------
+// Batched (multi-tensor) step: one launch updates every parameter of an
+// optimizer group. A transformer optimizer step is hundreds of tensors,
+// most of them small — per-tensor launches cost ~5-10 us each, which for
+// a 70B model's ~723 parameters is milliseconds of pure launch overhead.
+// Virtual blocks of kBatchEPB elements are assigned to tensors via a
+// prefix-sum table; each real block grid-strides over virtual blocks and
+// binary-searches its tensor.
+// ---------------------------------------------------------------------------
+
+constexpr int kBatchEPB = 2048;  // elements per virtual block (256 thr x 8)
+
+struct BatchTensorMeta {
+  void* param;
+  const void* grad;
+  void* exp_avg;
+  void* exp_avg_sq;
+  void* compensation;  // nullptr -> plain update
+  int64_t n;
+  int32_t dt_param, dt_grad, dt_m, dt_v, dt_c;
+  float step_size;
+  float bias_correction2_sqrt;
+};
+
+struct BatchHeader {
+  int32_t n_tensors;
+  int64_t total_vblocks;
+  float lr, beta1, beta2, eps, weight_decay;
+};
+
+// blob layout: int64 prefix[n_tensors] then BatchTensorMeta[n_tensors].
+__global__ void adamw_batched_kernel(const uint8_t* __restrict__ blob,
+                                     BatchHeader h) {
+  const int64_t* prefix = reinterpret_cast<const int64_t*>(blob);
+  const BatchTensorMeta* metas = reinterpret_cast<const BatchTensorMeta*>(
+      blob + sizeof(int64_t) * h.n_tensors);
+
+  for (int64_t vb = blockIdx.x; vb < h.total_vblocks; vb += gridDim.x) {
+    // Largest t with prefix[t] <= vb.
+    int lo = 0, hi = h.n_tensors - 1;
+    while (lo < hi) {
+      const int mid = (lo + hi + 1) >> 1;
+      if (prefix[mid] <= vb) {
+        lo = mid;
+      } else {
+        hi = mid - 1;
+      }
+    }
+    const BatchTensorMeta mt = metas[lo];
+    const int64_t base = (vb - prefix[lo]) * kBatchEPB;
+    const bool kahan = mt.compensation != nullptr;
+#pragma unroll
+    for (int j = 0; j < kBatchEPB / 256; ++j) {
+      const int64_t i = base + j * 256 + threadIdx.x;
+      if (i >= mt.n) {
+        break;
+      }
+      float p = loadf(mt.param, static_cast<DT>(mt.dt_param), i);
+      const float g = loadf(mt.grad, static_cast<DT>(mt.dt_grad), i);
+      float m = loadf(mt.exp_avg, static_cast<DT>(mt.dt_m), i);
+      float v = loadf(mt.exp_avg_sq, static_cast<DT>(mt.dt_v), i);
+
+      if (h.weight_decay != 0.0f) {
+        p *= 1.0f - h.lr * h.weight_decay;
+      }
+      m += (1.0f - h.beta1) * (g - m);
+      v = v * h.beta2 + g * g * (1.0f - h.beta2);
+      const float denom = sqrtf(v) / mt.bias_correction2_sqrt + h.eps;
+      const float update = -mt.step_size * (m / denom);
+
+      if (kahan) {
+        float c = loadf(mt.compensation, static_cast<DT>(mt.dt_c), i) +
+                  update;
+        const float prev = p;
+        storef(mt.param, static_cast<DT>(mt.dt_param), i, p + c);
+        const float p_stored =
+            loadf(mt.param, static_cast<DT>(mt.dt_param), i);
+        c += prev - p_stored;
+        storef(mt.compensation, static_cast<DT>(mt.dt_c), i, c);
+      } else {
+        p += update;
+        storef(mt.param, static_cast<DT>(mt.dt_param), i, p);
+      }
+      storef(mt.exp_avg, static_cast<DT>(mt.dt_m), i, m);
+      storef(mt.exp_avg_sq, static_cast<DT>(mt.dt_v), i, v);
+    }
+  }
+}
+
 DT toDT(const at::Tensor& t) {
   switch (t.scalar_type()) {
     case at::kFloat:
@@ -386,6 +475,95 @@ void anyprecision_adamw_step(at::Tensor& param,
                        stream.stream(), args);
     C10_HIP_KERNEL_LAUNCH_CHECK();
   }
+}
+
+void anyprecision_adamw_batched_step(
+    std::vector<at::Tensor> params,
+    std::vector<at::Tensor> grads,
+    std::vector<at::Tensor> exp_avgs,
+    std::vector<at::Tensor> exp_avg_sqs,
+    std::vector<std::optional<at::Tensor>> compensations,
+    double lr,
+    double beta1,
+    double beta2,
+    double eps,
+    double weight_decay,
+    std::vector<double> step_sizes,
+    std::vector<double> bias_correction2_sqrts) {
+  const size_t n_tensors = params.size();
+  TORCH_CHECK(n_tensors > 0, "batched AdamW: empty tensor list");
+  TORCH_CHECK(grads.size() == n_tensors && exp_avgs.size() == n_tensors &&
+                  exp_avg_sqs.size() == n_tensors &&
+                  compensations.size() == n_tensors &&
+                  step_sizes.size() == n_tensors &&
+                  bias_correction2_sqrts.size() == n_tensors,
+              "batched AdamW: list length mismatch");
+
+  const size_t prefix_bytes = sizeof(int64_t) * n_tensors;
+  const size_t blob_bytes =
+      prefix_bytes + sizeof(BatchTensorMeta) * n_tensors;
+  at::Tensor host_blob = at::empty(
+      {static_cast<int64_t>(blob_bytes)},
+      at::TensorOptions().dtype(at::kByte).pinned_memory(true));
+  auto* prefix = reinterpret_cast<int64_t*>(host_blob.data_ptr());
+  auto* metas = reinterpret_cast<BatchTensorMeta*>(
+      static_cast<uint8_t*>(host_blob.data_ptr()) + prefix_bytes);
+
+  int64_t total_vblocks = 0;
+  for (size_t t = 0; t < n_tensors; ++t) {
+    const at::Tensor& p = params[t];
+    const at::Tensor* checked[] = {&p, &grads[t], &exp_avgs[t],
+                                   &exp_avg_sqs[t]};
+    for (const at::Tensor* x : checked) {
+      TORCH_CHECK(x->is_cuda() && x->is_contiguous() &&
+                      x->unsafeGetTensorImpl()->has_storage() &&
+                      x->const_data_ptr() != nullptr &&
+                      x->numel() == p.numel(),
+                  "batched AdamW: tensors must be plain dense contiguous "
+                  "GPU tensors of equal numel");
+    }
+    BatchTensorMeta& mt = metas[t];
+    mt.param = p.data_ptr();
+    mt.grad = grads[t].data_ptr();
+    mt.exp_avg = exp_avgs[t].data_ptr();
+    mt.exp_avg_sq = exp_avg_sqs[t].data_ptr();
+    mt.compensation =
+        compensations[t].has_value() ? compensations[t]->data_ptr() : nullptr;
+    mt.n = p.numel();
+    mt.dt_param = static_cast<int32_t>(toDT(p));
+    mt.dt_grad = static_cast<int32_t>(toDT(grads[t]));
+    mt.dt_m = static_cast<int32_t>(toDT(exp_avgs[t]));
+    mt.dt_v = static_cast<int32_t>(toDT(exp_avg_sqs[t]));
+    mt.dt_c = compensations[t].has_value()
+                  ? static_cast<int32_t>(toDT(*compensations[t]))
+                  : static_cast<int32_t>(DT::kF32);
+    mt.step_size = static_cast<float>(step_sizes[t]);
+    mt.bias_correction2_sqrt =
+        static_cast<float>(bias_correction2_sqrts[t]);
+    prefix[t] = total_vblocks;
+    total_vblocks += (mt.n + kBatchEPB - 1) / kBatchEPB;
+  }
+
+  at::Tensor dev_blob =
+      host_blob.to(params[0].device(), /*non_blocking=*/true);
+
+  BatchHeader h;
+  h.n_tensors = static_cast<int32_t>(n_tensors);
+  h.total_vblocks = total_vblocks;
+  h.lr = static_cast<float>(lr);
+  h.beta1 = static_cast<float>(beta1);
+  h.beta2 = static_cast<float>(beta2);
+  h.eps = static_cast<float>(eps);
+  h.weight_decay = static_cast<float>(weight_decay);
+
+  const int grid =
+      static_cast<int>(std::min<int64_t>(total_vblocks, 16384));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(adamw_batched_kernel, dim3(grid), dim3(256), 0,
+                     stream.stream(),
+                     static_cast<const uint8_t*>(dev_blob.const_data_ptr()),
+                     h);
+  C10_HIP_KERNEL_LAUNCH_CHECK();
 }
 
 }  // namespace tdx

@@ -20,6 +20,20 @@ void anyprecision_adamw_step(at::Tensor& param,
                              double step_size,
                              double bias_correction2_sqrt);
 
+void anyprecision_adamw_batched_step(
+    std::vector<at::Tensor> params,
+    std::vector<at::Tensor> grads,
+    std::vector<at::Tensor> exp_avgs,
+    std::vector<at::Tensor> exp_avg_sqs,
+    std::vector<std::optional<at::Tensor>> compensations,
+    double lr,
+    double beta1,
+    double beta2,
+    double eps,
+    double weight_decay,
+    std::vector<double> step_sizes,
+    std::vector<double> bias_correction2_sqrts);
+
 }  // namespace tdx
 
 PYBIND11_MODULE(_K, m) {
@@ -27,6 +41,18 @@ PYBIND11_MODULE(_K, m) {
 
   m.def("has_init_kernels", [] { return true; });
   m.def("has_anyprecision_adamw", [] { return true; });
+  m.def("has_anyprecision_adamw_batched", [] { return true; });
+
+  m.def("anyprecision_adamw_batched_",
+        &tdx::anyprecision_adamw_batched_step, pybind11::arg("params"),
+        pybind11::arg("grads"), pybind11::arg("exp_avgs"),
+        pybind11::arg("exp_avg_sqs"), pybind11::arg("compensations"),
+        pybind11::arg("lr"), pybind11::arg("beta1"), pybind11::arg("beta2"),
+        pybind11::arg("eps"), pybind11::arg("weight_decay"),
+        pybind11::arg("step_sizes"),
+        pybind11::arg("bias_correction2_sqrts"),
+        "One-launch AdamW update for a whole list of parameters (kahan "
+        "optional per tensor via the compensations list).");
 
   m.def(
       "anyprecision_adamw_",

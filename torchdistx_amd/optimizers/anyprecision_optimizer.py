@@ -34,6 +34,11 @@ def _local_view(t):
     return t
 
 
+# Fused-eligible tensors below this element count batch into one kernel
+# launch per group; larger ones keep the per-tensor vectorized kernel.
+_BATCH_MAX_NUMEL = 1 << 20
+
+
 def _kernel_loaded() -> bool:
     try:
         from torchdistx_amd import _kernels
@@ -152,6 +157,13 @@ class AnyPrecisionAdamW(Optimizer):
             variance_dtype = group["variance_dtype"]
             compensation_dtype = group["compensation_buffer_dtype"]
 
+            # Fused-eligible small tensors accumulate here and update in
+            # ONE batched kernel launch at the end of the group — a
+            # transformer group is hundreds of tensors, and per-tensor
+            # launch overhead would otherwise dominate the small ones.
+            batch = {k: [] for k in
+                     ("p", "g", "m", "v", "c", "step_size", "bc2")}
+
             for p in group["params"]:
                 if p.grad is None:
                     continue
@@ -214,20 +226,31 @@ class AnyPrecisionAdamW(Optimizer):
                     from torchdistx_amd import _kernels
 
                     self._fused_steps += 1
-                    _kernels.anyprecision_adamw_(
-                        p_l,
-                        g_l,
-                        m_l,
-                        v_l,
-                        c_l,
-                        lr,
-                        beta1,
-                        beta2,
-                        eps,
-                        weight_decay,
-                        step_size,
-                        bias_correction2_sqrt,
-                    )
+                    if p_l.numel() >= _BATCH_MAX_NUMEL:
+                        # Big tensors take the per-tensor vectorized
+                        # kernel (16-byte packed accesses).
+                        _kernels.anyprecision_adamw_(
+                            p_l,
+                            g_l,
+                            m_l,
+                            v_l,
+                            c_l,
+                            lr,
+                            beta1,
+                            beta2,
+                            eps,
+                            weight_decay,
+                            step_size,
+                            bias_correction2_sqrt,
+                        )
+                    else:
+                        batch["p"].append(p_l)
+                        batch["g"].append(g_l)
+                        batch["m"].append(m_l)
+                        batch["v"].append(v_l)
+                        batch["c"].append(c_l)
+                        batch["step_size"].append(step_size)
+                        batch["bc2"].append(bias_correction2_sqrt)
                     continue
 
                 # Decoupled weight decay (AdamW).
@@ -261,3 +284,21 @@ class AnyPrecisionAdamW(Optimizer):
                     p.addcdiv_(
                         exp_avg.to(p.dtype), denom.to(p.dtype), value=-step_size
                     )
+
+            if batch["p"]:
+                from torchdistx_amd import _kernels
+
+                _kernels.anyprecision_adamw_batched_(
+                    batch["p"],
+                    batch["g"],
+                    batch["m"],
+                    batch["v"],
+                    batch["c"],
+                    lr,
+                    beta1,
+                    beta2,
+                    eps,
+                    weight_decay,
+                    batch["step_size"],
+                    batch["bc2"],
+                )
