@@ -18,6 +18,21 @@ from torchdistx_amd.deferred_init import deferred_init, materialize_tensor
 n_seeds = int(sys.argv[1]) if len(sys.argv) > 1 else 1000
 device = sys.argv[2] if len(sys.argv) > 2 else "cpu"
 seed_base = int(sys.argv[3]) if len(sys.argv) > 3 else 0
+threaded = len(sys.argv) > 4 and sys.argv[4] == "threaded"
+if threaded:
+    from test_tape_fuzz import _check_seed_threaded
+
+    fails = 0
+    for seed in range(seed_base, seed_base + n_seeds):
+        try:
+            _check_seed_threaded(seed, device)
+        except Exception as e:
+            fails += 1
+            print("FAIL", seed, repr(e)[:300])
+        if (seed - seed_base + 1) % 500 == 0:
+            print(f"{seed - seed_base + 1}/{n_seeds} threaded seeds, {fails} fails", flush=True)
+    print(f"done: {fails} fails over {n_seeds} threaded seeds on {device}")
+    sys.exit(1 if fails else 0)
 
 fails = 0
 for seed in range(seed_base, seed_base + n_seeds):

@@ -34,19 +34,42 @@ _DTYPES = [torch.float32, torch.bfloat16, torch.float16]
 
 
 def _random_chain(rng, shape, dtype, device):
-    kind = rng.choice(["uniform", "normal", "fill", "zero", "plain"])
+    kind = rng.choice(
+        ["uniform", "normal", "bernoulli", "fill", "zero", "plain"]
+    )
     t = torch.empty(shape, dtype=dtype, device=device)
     if kind == "uniform":
         a = rng.uniform(-2.0, 0.0)
         t.uniform_(a, a + rng.uniform(0.1, 3.0))
     elif kind == "normal":
         t.normal_(rng.uniform(-1.0, 1.0), rng.uniform(0.1, 2.0))
+    elif kind == "bernoulli":
+        t.bernoulli_(rng.uniform(0.0, 1.0))
     elif kind == "fill":
         t.fill_(rng.uniform(-3.0, 3.0))
     elif kind == "zero":
         t.zero_()
     else:  # plain factory via zeros (empty has no defined bits)
         t = torch.zeros(shape, dtype=dtype, device=device)
+    # Pointwise-scalar tail ops (trunc_normal_-style chains): slicing
+    # must commute with any mix of them.
+    for _ in range(rng.randint(0, 3)):
+        op = rng.choice(["mul", "add", "clamp", "abs", "erfinv", "neg"])
+        if op == "mul":
+            t.mul_(rng.uniform(-2.0, 2.0))
+        elif op == "add":
+            t.add_(rng.uniform(-1.0, 1.0))
+        elif op == "clamp":
+            lo = rng.uniform(-1.0, 0.0)
+            t.clamp_(min=lo, max=lo + rng.uniform(0.1, 2.0))
+        elif op == "abs":
+            t.abs_()
+        elif op == "erfinv":
+            # keep the domain in (-1, 1) first
+            t.clamp_(min=-0.999, max=0.999)
+            t.erfinv_()
+        else:
+            t.neg_()
     if rng.random() < 0.3:
         t = t.detach()
     return t

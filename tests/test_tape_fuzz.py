@@ -166,3 +166,62 @@ def test_random_program_replay_matches_eager_gpu_native(seed) -> None:
     if not torch.cuda.is_available():
         pytest.skip("needs a GPU")
     _check_seed(seed + 10_000, "cuda")
+
+
+def _check_seed_threaded(seed, device, n_threads=8, dtype=torch.float32):
+    # Same differential check, but the surviving tensors materialize from
+    # n_threads concurrent workers: overlapping alias families force the
+    # parallel replay machinery through its shared-node wait paths (the
+    # programs are RNG-free, so results are thread-order-independent).
+    import threading
+
+    rng = random.Random(seed)
+    ops = _build_program(rng, n_steps=25)
+    eager = _run_program(ops, device, dtype)
+
+    class Holder(Module):
+        def __init__(self):
+            super().__init__()
+            self.tensors = _run_program(ops, device, dtype)
+
+    holder = deferred_init(Holder)
+
+    order = list(range(len(eager)))
+    rng.shuffle(order)
+    results = {}
+    errors = []
+    lock = threading.Lock()
+
+    def worker(chunk):
+        try:
+            for i in chunk:
+                got = materialize_tensor(holder.tensors[i])
+                with lock:
+                    results[i] = got
+        except Exception as e:  # pragma: no cover
+            errors.append(e)
+
+    chunks = [order[k::n_threads] for k in range(n_threads)]
+    threads = [
+        threading.Thread(target=worker, args=(c,)) for c in chunks if c
+    ]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert not errors, (seed, errors)
+    for i in order:
+        assert torch.equal(results[i], eager[i]), (seed, i, ops)
+
+
+@pytest.mark.parametrize("seed", range(25))
+def test_random_program_threaded_replay_matches_eager(seed) -> None:
+    _check_seed_threaded(seed + 70_000, "cpu")
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("seed", range(15))
+def test_random_program_threaded_replay_gpu(seed) -> None:
+    if not torch.cuda.is_available():
+        pytest.skip("needs a ROCm GPU")
+    _check_seed_threaded(seed + 90_000, "cuda")
