@@ -403,11 +403,16 @@ def test_batched_anyprecision_adamw_matches_per_tensor() -> None:
         assert torch.allclose(pr, pb, rtol=1e-6, atol=1e-7)
         assert torch.allclose(mr, mb, rtol=1e-6, atol=1e-7)
         assert torch.allclose(vr, vb, rtol=1e-6, atol=1e-7)
-        # The vec and batched kernels round the Kahan remainder through
-        # slightly different instruction sequences (explicit convert vs
-        # store-reload); the compensation may differ in the last bf16 ulp.
+        # The compensation buffer alone is a catastrophic-cancellation
+        # residual (prev - rounded(p+c)): a single-ulp difference in how
+        # the vec vs batched instruction sequences round p propagates
+        # into c at arbitrary relative size. The meaningful quantity is
+        # the effective parameter p + c, which must agree tightly.
         assert torch.allclose(
-            cr.float(), cb.float(), rtol=1e-2, atol=1e-9
+            pr.float() + cr.float(),
+            pb.float() + cb.float(),
+            rtol=1e-6,
+            atol=1e-7,
         )
 
 
