@@ -31,6 +31,7 @@
 #include <ATen/CPUGeneratorImpl.h>
 #include <ATen/Context.h>
 #include <ATen/ThreadLocalState.h>
+#include <c10/core/GradMode.h>
 #include <ATen/core/dispatch/Dispatcher.h>
 #include <ATen/core/VariableHooksInterface.h>
 #include <c10/core/DefaultDtype.h>
@@ -192,8 +193,13 @@ void beginRngSession() {
 }
 
 void endRngSession() {
+  // Registry release and the exit-advance must be ONE atomic step with
+  // respect to other threads opening segments: otherwise a concurrent
+  // thread can snapshot the pre-advance generator state, lose the race
+  // to this release, and register the same state hash with salt 0 —
+  // pinning the stream this session already used.
+  std::lock_guard<std::mutex> reg_lock(segment_registry_mutex);
   if (current_rng_session != nullptr) {
-    std::lock_guard<std::mutex> lock(segment_registry_mutex);
     for (uint64_t h : current_rng_session->raw_hashes) {
       auto it = live_segment_uses.find(h);
       if (it != live_segment_uses.end() && --(it->second) == 0) {
@@ -221,6 +227,13 @@ std::optional<std::pair<uint64_t, uint64_t>> pinPhiloxForOp(
     return std::nullopt;
   }
   RngSession& session = *current_rng_session;
+  // The registry lock brackets the state snapshot AND the segment
+  // registration, and session exit advances the generator under the
+  // same lock — so "state hash is live in the registry" and "state has
+  // been advanced past" are mutually exclusive, with no window where a
+  // second thread can re-pin a just-closed session's stream (gen.mutex
+  // nests inside it, same order as endRngSession).
+  std::lock_guard<std::mutex> reg_lock(segment_registry_mutex);
   auto gen = at::globalContext().defaultGenerator(c10::DeviceType::CPU);
   at::Tensor state;
   {
@@ -233,11 +246,7 @@ std::optional<std::pair<uint64_t, uint64_t>> pinPhiloxForOp(
     session.has_segment = true;
     session.segment_hash = h;
     session.next_slot = 1;
-    uint64_t salt = 0;
-    {
-      std::lock_guard<std::mutex> lock(segment_registry_mutex);
-      salt = live_segment_uses[h]++;
-    }
+    const uint64_t salt = live_segment_uses[h]++;
     session.raw_hashes.push_back(h);
     session.nonce = salt == 0 ? h : mixSalt(h, salt);
     if (salt != 0) {
@@ -525,7 +534,12 @@ void deferredInitHandler(const c10::OperatorHandle& op,
   if (has_fake_arg && isInPlaceInitOp(schema.operator_name()) &&
       (*stack)[args_begin].isTensor()) {
     const at::Tensor& self = (*stack)[args_begin].toTensor();
-    if (isFake(self)) {
+    // Autograd-visible mutations (in-place on a requires_grad tensor
+    // under grad mode) must keep going through the autograd layer so
+    // eager's semantics — the leaf-mutation error, version bumps —
+    // surface at record time exactly as they would eagerly.
+    if (isFake(self) &&
+        (!self.requires_grad() || !c10::GradMode::is_enabled())) {
       const auto& opname = schema.operator_name().name;
       if (opname == "aten::uniform_" || opname == "aten::normal_") {
         TORCH_CHECK(c10::isFloatingType(self.scalar_type()), "`", opname,
@@ -745,6 +759,10 @@ class ProxyVariableHooks final : public at::impl::VariableHooksInterface {
     return inner_;
   }
 
+  void setInner(at::impl::VariableHooksInterface* inner) noexcept {
+    inner_ = inner;
+  }
+
  private:
   at::impl::VariableHooksInterface* inner_;
 };
@@ -752,13 +770,26 @@ class ProxyVariableHooks final : public at::impl::VariableHooksInterface {
 std::mutex hooks_mutex;
 size_t hooks_refcount = 0;
 at::impl::VariableHooksInterface* saved_hooks = nullptr;
+// Never freed: a thread that fetched the global hooks pointer just
+// before the last session left (e.g. replaying a recorded
+// tdx::variable_data op, which calls through the hooks) may still be
+// mid-call when the stock hooks are restored — deleting the proxy there
+// is a use-after-free. The proxy forwards to the real hooks and records
+// nothing outside an active session, so keeping the one object alive
+// for the process lifetime is both safe and semantically inert.
 ProxyVariableHooks* proxy_hooks = nullptr;
 
 void installProxyHooks() {
   std::lock_guard<std::mutex> lock{hooks_mutex};
   if (hooks_refcount++ == 0) {
-    saved_hooks = at::impl::GetVariableHooks();
-    proxy_hooks = new ProxyVariableHooks{saved_hooks};
+    at::impl::VariableHooksInterface* current = at::impl::GetVariableHooks();
+    TORCH_INTERNAL_ASSERT(current != proxy_hooks);
+    saved_hooks = current;
+    if (proxy_hooks == nullptr) {
+      proxy_hooks = new ProxyVariableHooks{saved_hooks};
+    } else {
+      proxy_hooks->setInner(saved_hooks);
+    }
     at::impl::SetVariableHooks(proxy_hooks);
   }
 }
@@ -768,9 +799,6 @@ void removeProxyHooks() {
   TORCH_INTERNAL_ASSERT(hooks_refcount > 0);
   if (--hooks_refcount == 0) {
     at::impl::SetVariableHooks(saved_hooks);
-    delete proxy_hooks;
-    proxy_hooks = nullptr;
-    saved_hooks = nullptr;
   }
 }
 
