@@ -145,3 +145,48 @@ def test_parallel_replay_speedup() -> None:
         _C.set_native_init_cpu(False)
     speedup = serial / parallel
     assert speedup > 1.6, f"parallel replay speedup only {speedup:.2f}x"
+
+
+def test_concurrent_sessions_stress() -> None:
+    # Regression for two races found under stress: the segment-nonce
+    # TOCTOU (concurrent unseeded sessions re-pinning one stream) and
+    # the VariableHooks-proxy use-after-free on session exit. 40
+    # iterations of two fully-overlapping record+materialize sessions;
+    # any recurrence shows up as equal weights or a crash.
+    import torch as _torch
+
+    from torchdistx_amd import _C
+    from torchdistx_amd.deferred_init import (
+        deferred_init as _dinit,
+        materialize_module as _mat,
+    )
+
+    def one():
+        m = _dinit(lambda: nn.Linear(16, 16))
+        _C.set_native_init_cpu(True)
+        try:
+            _mat(m)
+        finally:
+            _C.set_native_init_cpu(False)
+        return m
+
+    for trial in range(40):
+        _torch.manual_seed(2024)
+        results = {}
+        barrier = threading.Barrier(2)
+        errors = []
+
+        def build(key):
+            try:
+                barrier.wait()
+                results[key] = one()
+            except Exception as e:  # pragma: no cover
+                errors.append(e)
+
+        ts = [threading.Thread(target=build, args=(i,)) for i in range(2)]
+        for t in ts:
+            t.start()
+        for t in ts:
+            t.join()
+        assert not errors, errors
+        assert not torch.equal(results[0].weight, results[1].weight), trial
