@@ -21,6 +21,7 @@
 #include <algorithm>
 #include <atomic>
 #include <condition_variable>
+#include <cstring>
 #include <functional>
 #include <memory>
 #include <mutex>
@@ -167,24 +168,27 @@ bool isRngOpName(const std::string& name) {
          name == "aten::bernoulli_";
 }
 
-// FNV-1a over a generator-state blob.
+// Word-at-a-time mix over a generator-state blob (splitmix-style):
+// change detection + nonce derivation, not cryptography. Runs once per
+// recorded RNG op over the ~5 KB Mersenne state, so byte-at-a-time FNV
+// (~5 us per op) would be a measurable recording cost.
 uint64_t hashStateBytes(const uint8_t* bytes, int64_t n) {
-  uint64_t h = 1469598103934665603ull;
-  uint64_t word = 0;
-  for (int64_t i = 0; i < n; ++i) {
-    word = (word << 8) | bytes[i];
-    if ((i & 7) == 7) {
-      for (int b = 0; b < 8; ++b) {
-        h = (h ^ ((word >> (8 * b)) & 0xff)) * 1099511628211ull;
-      }
-      word = 0;
-    }
+  uint64_t h = 0x9E3779B97F4A7C15ull ^ static_cast<uint64_t>(n);
+  int64_t i = 0;
+  for (; i + 8 <= n; i += 8) {
+    uint64_t w;
+    std::memcpy(&w, bytes + i, 8);
+    h ^= w;
+    h *= 0xBF58476D1CE4E5B9ull;
+    h ^= h >> 29;
   }
-  if ((n & 7) != 0) {
-    for (int b = 0; b < 8; ++b) {
-      h = (h ^ ((word >> (8 * b)) & 0xff)) * 1099511628211ull;
-    }
+  uint64_t tail = 0;
+  for (int j = 0; i < n; ++i, ++j) {
+    tail |= static_cast<uint64_t>(bytes[i]) << (8 * j);
   }
+  h ^= tail;
+  h *= 0x94D049BB133111EBull;
+  h ^= h >> 32;
   return h;
 }
 
