@@ -447,3 +447,31 @@ def test_optimizer_batches_small_tensors() -> None:
     assert opt._fused_steps == 3 * len(params)
     for p, e in zip(params, eager_params):
         assert torch.allclose(p, e, rtol=1e-5, atol=1e-6)
+
+
+def test_fp8_quantized_init_through_deferred() -> None:
+    # fp8 init-time quantization: ATen has no fp8 RNG kernels, so the
+    # idiomatic MI355X flow records a bf16 init followed by a cast —
+    # replay regenerates the bf16 master bits through the tdx Philox
+    # kernels and quantizes to OCP fp8 on-device.
+    from torchdistx_amd import deferred_init
+    from torchdistx_amd.deferred_init import materialize_module
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            w = torch.empty(256, 128, device="cuda", dtype=torch.bfloat16)
+            w.normal_(0.0, 0.05)
+            self.register_buffer("w8", w.to(torch.float8_e4m3fn))
+
+    torch.manual_seed(12)
+    m = deferred_init(M)
+    materialize_module(m)
+    assert m.w8.dtype == torch.float8_e4m3fn and m.w8.is_cuda
+
+    torch.manual_seed(12)
+    ref = deferred_init(M)
+    materialize_module(ref)
+    assert torch.equal(m.w8.view(torch.uint8), ref.w8.view(torch.uint8))
+    f = m.w8.float()
+    assert f.std().item() == pytest.approx(0.05, rel=0.1)
