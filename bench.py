@@ -54,6 +54,11 @@ def get_args():
                         "allocator cross-stream overhead (measured t1 "
                         "53.3 ms vs t4 67.7 ms); the parallel path pays "
                         "off for many-small-tensor CPU/mixed workloads.")
+    p.add_argument("--replay", type=str, default="pertensor",
+                   choices=["pertensor", "batched"],
+                   help="batched: collapse simple init chains and fill all "
+                        "tensors in one kernel launch (bitwise-identical "
+                        "to per-tensor replay)")
     p.add_argument("--selftest", action="store_true",
                    help="multi-rank preflight: validates communicator "
                         "creation, cross-rank bitwise equality of every "
@@ -142,6 +147,13 @@ def main():
             shards = materialize_module_dim0_sharded(module, rank, world)
             assert shards
             del shards
+        elif args.mode == "replicate" and use_cuda and args.replay == "batched":
+            from torchdistx_amd.deferred_init import (
+                materialize_module_batched,
+            )
+
+            materialize_module_batched(module)
+            assert not is_deferred(module), "materialization incomplete"
         elif args.mode == "replicate" and use_cuda and args.threads > 1:
             from torchdistx_amd.deferred_init import (
                 materialize_module_parallel,
@@ -214,6 +226,7 @@ def main():
                 "mode": args.mode,
                 "init": args.init,
                 "threads": args.threads,
+                "replay": args.replay,
                 "device": "cuda" if use_cuda else "cpu",
                 "native_init_kernels": _kernels.available(),
                 "peak_host_rss_gb": round(rss_gb, 2),

@@ -614,3 +614,45 @@ def test_partial_checkpoint_load_then_materialize_rest() -> None:
         assert torch.equal(m.b.bias, full.b.bias)
     finally:
         _C.set_native_init_cpu(False)
+
+
+def test_tensor_init_plan_extraction() -> None:
+    # Simple chains reduce to their final value step; pointwise tails
+    # and cross-tensor dependencies return None (fallback to replay).
+    from torchdistx_amd import _C
+
+    class M(Module):
+        def __init__(self):
+            super().__init__()
+            self.n = Parameter(torch.empty(8, 4).normal_(0.5, 2.0))
+            self.z = Parameter(torch.zeros(6))
+            w = torch.empty(4, 4)
+            torch.nn.init.trunc_normal_(w)
+            self.t = Parameter(w)  # pointwise tail -> not plannable
+            self.d = Parameter(torch.zeros(3) + torch.ones(3))  # dep
+
+    m = deferred_init(M)
+    plan = _C.tensor_init_plan(m.n)
+    assert plan is not None
+    assert plan["kind"] == "normal"
+    assert plan["p0"] == 0.5 and plan["p1"] == 2.0
+    assert plan["sizes"] == [8, 4]
+    assert plan["seed"] != 0
+    z = _C.tensor_init_plan(m.z)
+    assert z is not None and z["kind"] == "zero"
+    assert _C.tensor_init_plan(m.t) is None
+    assert _C.tensor_init_plan(m.d) is None
+    # planning must not consume the tape:
+    materialize_module(m)
+    assert not is_deferred(m)
+
+
+def test_materialize_module_batched_cpu_fallback() -> None:
+    from torchdistx_amd.deferred_init import materialize_module_batched
+
+    torch.manual_seed(88)
+    m = deferred_init(torch.nn.Linear, 8, 8)
+    materialize_module_batched(m)
+    torch.manual_seed(88)
+    e = torch.nn.Linear(8, 8)
+    assert torch.equal(m.weight, e.weight)

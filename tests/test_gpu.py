@@ -475,3 +475,35 @@ def test_fp8_quantized_init_through_deferred() -> None:
     assert torch.equal(m.w8.view(torch.uint8), ref.w8.view(torch.uint8))
     f = m.w8.float()
     assert f.std().item() == pytest.approx(0.05, rel=0.1)
+
+
+def test_batched_replay_bitwise_equal() -> None:
+    # The batched replay planner (one launch for all simple init chains)
+    # must be bitwise-identical to per-tensor replay: same per-tensor
+    # Philox streams, same group indexing.
+    from torchdistx_amd import deferred_init
+    from torchdistx_amd.deferred_init import (
+        materialize_module,
+        materialize_module_batched,
+    )
+    from torchdistx_amd.models import TINY, build_model
+
+    torch.manual_seed(41)
+    seq = deferred_init(build_model, TINY, device="cuda", dtype=torch.bfloat16)
+    materialize_module(seq)
+
+    torch.manual_seed(41)
+    bat = deferred_init(build_model, TINY, device="cuda", dtype=torch.bfloat16)
+    materialize_module_batched(bat)
+    torch.cuda.synchronize()
+
+    for (n1, p1), (n2, p2) in zip(
+        seq.named_parameters(), bat.named_parameters()
+    ):
+        assert n1 == n2 and torch.equal(p1, p2), n1
+    for (n1, b1), (n2, b2) in zip(seq.named_buffers(), bat.named_buffers()):
+        assert n1 == n2 and torch.equal(b1, b2), n1
+    assert not any(
+        p.requires_grad is False for p in bat.parameters()
+        if p.requires_grad != p.requires_grad
+    )

@@ -1328,6 +1328,80 @@ void applyShardStep(const ChainStep& step, at::Tensor& shard, int64_t start,
 
 }  // namespace
 
+// Reduces a deferred tensor's tape chain to its final whole-tensor value
+// step, when the chain is that simple. Every simple-chain value step
+// (uniform/normal/bernoulli/fill/zero) overwrites the whole tensor, so
+// only the LAST one determines the bits: batched replay can then fill
+// many tensors with ONE kernel launch. Returns nullopt for anything
+// else (pointwise tails, views, cross-tensor deps, explicit
+// generators); callers fall back to ordinary materialization.
+std::optional<InitPlan> tensorInitPlan(const at::Tensor& tensor) {
+  auto* fake = asFake(tensor);
+  if (fake == nullptr) {
+    return std::nullopt;
+  }
+  auto rec = getRecord(fake);
+  if (rec == nullptr) {
+    return std::nullopt;
+  }
+  const at::Tensor& meta = fake->meta_tensor();
+  if (!meta.is_contiguous()) {
+    return std::nullopt;
+  }
+  std::lock_guard<std::recursive_mutex> lock{tape_mutex};
+  InitPlan plan;
+  try {
+    auto nodes = buildCallStack(*rec);
+    bool have_value = false;
+    for (const auto& n : nodes) {
+      ChainStep step = classifyChainNode(*n);
+      switch (step.kind) {
+        case ChainStep::Kind::kFactory:
+        case ChainStep::Kind::kPass:
+          break;
+        case ChainStep::Kind::kPointwise:
+          return std::nullopt;  // needs sequential application
+        case ChainStep::Kind::kZero:
+          plan.kind = InitPlan::Kind::kZero;
+          have_value = true;
+          break;
+        case ChainStep::Kind::kFill:
+          plan.kind = InitPlan::Kind::kFill;
+          plan.p0 = step.p0;
+          have_value = true;
+          break;
+        case ChainStep::Kind::kUniform:
+        case ChainStep::Kind::kNormal:
+        case ChainStep::Kind::kBernoulli:
+          if (!step.philox.has_value()) {
+            return std::nullopt;
+          }
+          plan.kind = step.kind == ChainStep::Kind::kUniform
+                          ? InitPlan::Kind::kUniform
+                          : (step.kind == ChainStep::Kind::kNormal
+                                 ? InitPlan::Kind::kNormal
+                                 : InitPlan::Kind::kBernoulli);
+          plan.p0 = step.p0;
+          plan.p1 = step.p1;
+          plan.seed = step.philox->first;
+          plan.offset = step.philox->second;
+          have_value = true;
+          break;
+      }
+    }
+    if (!have_value) {
+      plan.kind = InitPlan::Kind::kFactory;  // allocate-only (empty)
+    }
+  } catch (const c10::Error&) {
+    return std::nullopt;  // not a simple chain
+  }
+  plan.sizes.assign(meta.sizes().begin(), meta.sizes().end());
+  plan.dtype = meta.scalar_type();
+  plan.device = fake->fake_device();
+  plan.requires_grad = tensor.requires_grad();
+  return plan;
+}
+
 at::Tensor materializeTensorShard(const at::Tensor& tensor,
                                   int64_t start_row,
                                   int64_t end_row) {

@@ -57,6 +57,22 @@ struct RecordInfo {
 // std::nullopt when `tensor` carries no deferred-init record.
 std::optional<RecordInfo> recordInfo(const at::Tensor& tensor);
 
+// Reduced replay plan of a simple init chain (see tensorInitPlan).
+struct InitPlan {
+  enum class Kind { kFactory, kUniform, kNormal, kBernoulli, kFill, kZero };
+  Kind kind = Kind::kFactory;
+  double p0 = 0.0;
+  double p1 = 0.0;
+  uint64_t seed = 0;
+  uint64_t offset = 0;
+  std::vector<int64_t> sizes;
+  c10::ScalarType dtype = c10::ScalarType::Float;
+  c10::Device device{c10::DeviceType::CPU};
+  bool requires_grad = false;
+};
+
+std::optional<InitPlan> tensorInitPlan(const at::Tensor& tensor);
+
 // Slice materialization: materializes rows [start_row, end_row) of the
 // deferred tensor's dim 0 WITHOUT materializing the rest, bitwise-equal to
 // the corresponding slice of a full materialization (per device type).

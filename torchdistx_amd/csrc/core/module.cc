@@ -112,6 +112,28 @@ PYBIND11_MODULE(_C, m) {
   }, "Replays the tensor's tape segment (GIL released); identity for real "
      "tensors, stable object across repeated calls and aliases.");
 
+  m.def("tensor_init_plan", [](const at::Tensor& t) -> pybind11::object {
+    auto plan = tdx::tensorInitPlan(t);
+    if (!plan.has_value()) {
+      return pybind11::none();
+    }
+    static const char* kKinds[] = {"factory", "uniform", "normal",
+                                   "bernoulli", "fill", "zero"};
+    pybind11::dict d;
+    d["kind"] = kKinds[static_cast<int>(plan->kind)];
+    d["p0"] = plan->p0;
+    d["p1"] = plan->p1;
+    d["seed"] = static_cast<int64_t>(plan->seed);
+    d["offset"] = static_cast<int64_t>(plan->offset);
+    d["sizes"] = plan->sizes;
+    d["dtype"] = pybind11::cast(plan->dtype);
+    d["device"] = pybind11::cast(plan->device);
+    d["requires_grad"] = plan->requires_grad;
+    return d;
+  }, "Reduced replay plan of a simple init chain (the final whole-tensor "
+     "value step), or None when the tape is not that simple. Feeds the "
+     "batched replay planner (materialize_module_batched).");
+
   m.def("materialize_tensor_shard",
         [](const at::Tensor& t, int64_t start_row, int64_t end_row) {
           at::Tensor out;

@@ -20,6 +20,13 @@ void anyprecision_adamw_step(at::Tensor& param,
                              double step_size,
                              double bias_correction2_sqrt);
 
+void batched_init_launch(std::vector<at::Tensor> tensors,
+                         std::vector<int64_t> dists,
+                         std::vector<double> p0s,
+                         std::vector<double> p1s,
+                         std::vector<int64_t> seeds,
+                         std::vector<int64_t> offsets);
+
 void anyprecision_adamw_batched_step(
     std::vector<at::Tensor> params,
     std::vector<at::Tensor> grads,
@@ -42,6 +49,15 @@ PYBIND11_MODULE(_K, m) {
   m.def("has_init_kernels", [] { return true; });
   m.def("has_anyprecision_adamw", [] { return true; });
   m.def("has_anyprecision_adamw_batched", [] { return true; });
+  m.def("has_batched_init", [] { return true; });
+
+  m.def("batched_init_", &tdx::batched_init_launch,
+        pybind11::arg("tensors"), pybind11::arg("dists"),
+        pybind11::arg("p0s"), pybind11::arg("p1s"), pybind11::arg("seeds"),
+        pybind11::arg("offsets"),
+        "One launch filling every listed tensor with its reduced init "
+        "plan (dist: 0 uniform, 1 normal, 2 bernoulli, 3 fill, 4 zero); "
+        "bitwise-identical to the per-tensor tdx kernels.");
 
   m.def("anyprecision_adamw_batched_",
         &tdx::anyprecision_adamw_batched_step, pybind11::arg("params"),
