@@ -18,6 +18,7 @@ _EXAMPLES = os.path.join(
         "train_fsdp_slowmo.py",
         "train_fsdp2_anyprecision.py",
         "init_405b_sharded.py",
+        "init_from_checkpoint.py",
     ],
 )
 def test_example_compiles(name: str) -> None:
