@@ -170,3 +170,29 @@ def test_fsdp2_anyprecision_optimizer_step(nccl_world1) -> None:
     n_params = sum(1 for p in module.parameters() if p.grad is not None)
     assert optim._fused_steps >= 3 * max(n_params - 1, 1), (
         optim._fused_steps, n_params)
+
+
+def test_shard_mode_batched_fill_bitwise(nccl_world1) -> None:
+    # Shard-mode owned tensors fill through the batched planner; at world
+    # 1 this rank owns everything and must match plain materialization
+    # bitwise.
+    from torchdistx_amd import deferred_init
+    from torchdistx_amd.deferred_init import materialize_module
+    from torchdistx_amd.models import TINY, build_model
+    from torchdistx_amd.parallel import materialize_module_distributed
+
+    torch.manual_seed(17)
+    ref = deferred_init(build_model, TINY, device="cuda",
+                        dtype=torch.bfloat16)
+    materialize_module(ref)
+
+    torch.manual_seed(17)
+    m = deferred_init(build_model, TINY, device="cuda",
+                      dtype=torch.bfloat16)
+    owner_map = materialize_module_distributed(m, mode="shard")
+    assert owner_map and all(o == 0 for o in owner_map.values())
+    torch.cuda.synchronize()
+    for (n1, p1), (n2, p2) in zip(
+        ref.named_parameters(), m.named_parameters()
+    ):
+        assert n1 == n2 and torch.equal(p1, p2), n1

@@ -189,6 +189,24 @@ def materialize_module_batched(
             if b is not None and _C.can_materialize(b):
                 entries.append((submodule, key, b, False))
 
+    for entry in _batched_fill(entries):
+        submodule, key, tensor, is_param = entry
+        mat = materialize_tensor(tensor)
+        if is_param:
+            submodule._parameters[key] = mat
+        else:
+            submodule._buffers[key] = mat
+
+
+def _batched_fill(entries) -> list:
+    """Fills every plannable (simple-chain, GPU, f32/bf16/f16) entry with
+    one batched kernel launch and swaps the module slots; returns the
+    entries that must go through ordinary replay instead. Used by
+    :func:`materialize_module_batched` and the distributed shard path."""
+    import torch
+
+    from torchdistx_amd import _kernels
+
     gpu = (
         bool(entries)
         and entries[0][2].is_cuda
@@ -196,8 +214,7 @@ def materialize_module_batched(
         and _kernels._K.has_batched_init()
     )
     if not gpu:
-        materialize_module(module, buffers_only, check_fn)
-        return
+        return list(entries)
 
     batch = {"t": [], "dist": [], "p0": [], "p1": [], "seed": [],
              "offset": [], "entry": []}
@@ -237,12 +254,7 @@ def materialize_module_batched(
             submodule._parameters[key] = out
         else:
             submodule._buffers[key] = out
-    for submodule, key, tensor, is_param in fallback:
-        mat = materialize_tensor(tensor)
-        if is_param:
-            submodule._parameters[key] = mat
-        else:
-            submodule._buffers[key] = mat
+    return fallback
 
 
 def materialize_module_parallel(

@@ -105,9 +105,12 @@ def materialize_module_distributed(
     owners = assign_owners([t.numel() for _, _, t, _ in entries], world)
 
     if mode == "shard":
-        for i, (submodule, key, tensor, is_param) in enumerate(entries):
-            if rank != owners[i]:
-                continue
+        from torchdistx_amd.deferred_init import _batched_fill
+
+        owned = [e for i, e in enumerate(entries) if owners[i] == rank]
+        # Simple-chain owned tensors fill in one batched launch; the
+        # rest replay through the tape.
+        for submodule, key, tensor, is_param in _batched_fill(owned):
             mat = _C.materialize_tensor(tensor)
             if is_param:
                 mat = _restore_class(tensor, mat)
