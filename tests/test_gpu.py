@@ -507,3 +507,30 @@ def test_batched_replay_bitwise_equal() -> None:
         p.requires_grad is False for p in bat.parameters()
         if p.requires_grad != p.requires_grad
     )
+
+
+def test_batched_replay_preserves_tied_parameters() -> None:
+    # A tied (shared-object) parameter must come out of the batched
+    # planner as ONE allocation referenced from both slots, exactly like
+    # tape replay's identity-stable materialization.
+    from torchdistx_amd import deferred_init
+    from torchdistx_amd.deferred_init import materialize_module_batched
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            p = torch.nn.Parameter(
+                torch.empty(32, 16, device="cuda").normal_(0, 0.1)
+            )
+            self.a = torch.nn.Linear(16, 32, bias=False, device="cuda")
+            self.b = torch.nn.Linear(16, 32, bias=False, device="cuda")
+            self.a.weight = p
+            self.b.weight = p
+
+    torch.manual_seed(61)
+    m = deferred_init(M)
+    assert m.a.weight is m.b.weight
+    materialize_module_batched(m)
+    torch.cuda.synchronize()
+    assert m.a.weight is m.b.weight
+    assert m.a.weight.is_cuda and m.a.weight.shape == (32, 16)

@@ -219,8 +219,12 @@ def _batched_fill(entries) -> list:
     batch = {"t": [], "dist": [], "p0": [], "p1": [], "seed": [],
              "offset": [], "entry": []}
     fallback = []
+    seen = {}  # id(fake) -> out: tied params must share ONE allocation
     for entry in entries:
         tensor = entry[2]
+        if id(tensor) in seen:
+            batch["entry"].append((entry, seen[id(tensor)]))
+            continue
         plan = _C.tensor_init_plan(tensor)
         if plan is None or plan["dtype"] not in (
             torch.float32, torch.bfloat16, torch.float16
@@ -230,6 +234,7 @@ def _batched_fill(entries) -> list:
         out = torch.empty(
             plan["sizes"], dtype=plan["dtype"], device=plan["device"]
         )
+        seen[id(tensor)] = out
         code = _DIST_CODE[plan["kind"]]
         if code >= 0:
             batch["t"].append(out)
