@@ -401,3 +401,78 @@ def test_bernoulli_stock_replay_matches_eager() -> None:
     torch.manual_seed(404)
     e = M()
     assert torch.equal(m.p, e.p)
+
+
+def test_broadcast_preserves_tied_parameters_multiproc() -> None:
+    # Tied parameters must land as ONE tensor object per rank after a
+    # bucketed broadcast (receivers included).
+    from tests._dist_utils import run_distributed
+
+    run_distributed(_tied_broadcast_worker, world_size=2)
+
+
+def _tied_broadcast_worker(rank, world_size):
+    import torch
+    from torch.nn import Linear, Parameter
+
+    from torchdistx_amd import _C, deferred_init
+    from torchdistx_amd.parallel import materialize_module_distributed
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            p = Parameter(torch.empty(8, 8).normal_())
+            self.a = Linear(8, 8, bias=False)
+            self.b = Linear(8, 8, bias=False)
+            self.a.weight = p
+            self.b.weight = p
+
+    torch.manual_seed(900 + rank)  # rank-skewed: only the wire reconciles
+    m = deferred_init(M)
+    assert m.a.weight is m.b.weight
+    _C.set_native_init_cpu(True)
+    try:
+        materialize_module_distributed(m, mode="broadcast")
+    finally:
+        _C.set_native_init_cpu(False)
+    assert m.a.weight is m.b.weight, "tied param split by broadcast"
+
+
+def test_allgather_and_shard_tied_params_multiproc() -> None:
+    from tests._dist_utils import run_distributed
+
+    run_distributed(_tied_shard_worker, world_size=2)
+
+
+def _tied_shard_worker(rank, world_size):
+    import torch
+    from torch.nn import Linear, Parameter
+
+    from torchdistx_amd import _C, deferred_init
+    from torchdistx_amd.parallel import materialize_module_distributed
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            p = Parameter(torch.empty(8, 8).normal_())
+            self.a = Linear(8, 8, bias=False)
+            self.b = Linear(8, 8, bias=False)
+            self.a.weight = p
+            self.b.weight = p
+
+    torch.manual_seed(901)
+    m = deferred_init(M)
+    _C.set_native_init_cpu(True)
+    try:
+        owner_map = materialize_module_distributed(m, mode="shard")
+    finally:
+        _C.set_native_init_cpu(False)
+    # One unique tensor -> one ownership entry; on the owner both slots
+    # hold the same object, elsewhere both stay fake.
+    assert len(owner_map) == 1
+    owner = owner_map[0]
+    if rank == owner:
+        assert m.a.weight is m.b.weight
+        assert not _C.can_materialize(m.a.weight)
+    else:
+        assert _C.can_materialize(m.a.weight)

@@ -55,6 +55,42 @@ def _named_deferred_tensors(
     return out
 
 
+def _dedupe_entries(entries):
+    """Tied parameters surface as several (module, key) slots holding ONE
+    fake object; distributed materialization must treat them as one
+    tensor (one owner, one transfer, one result object). Returns the
+    unique entries, the duplicate slots, and the id->unique-entry map."""
+    unique, dups, first = [], [], {}
+    for e in entries:
+        tid = id(e[2])
+        if tid in first:
+            dups.append(e)
+        else:
+            first[tid] = e
+            unique.append(e)
+    return unique, dups, first
+
+
+def _relink_duplicates(dups, first) -> None:
+    """Points every duplicate slot at whatever its unique counterpart's
+    slot now holds (if that slot was materialized on this rank)."""
+    from torchdistx_amd import _C as _core
+
+    for submodule, key, tensor, is_param in dups:
+        u_sub, u_key, _, u_is_param = first[id(tensor)]
+        val = (
+            u_sub._parameters[u_key]
+            if u_is_param
+            else u_sub._buffers[u_key]
+        )
+        if val is None or _core.can_materialize(val):
+            continue  # unique slot not materialized on this rank (shard)
+        if is_param:
+            submodule._parameters[key] = val
+        else:
+            submodule._buffers[key] = val
+
+
 def assign_owners(sizes: List[int], world_size: int) -> List[int]:
     """Greedy size-balanced deterministic assignment: largest tensors
     first, each to the currently least-loaded rank."""
@@ -102,6 +138,7 @@ def materialize_module_distributed(
     entries = _named_deferred_tensors(module, buffers_only)
     if check_fn is not None:
         entries = [e for e in entries if check_fn(e[0])]
+    entries, dups, first = _dedupe_entries(entries)
     owners = assign_owners([t.numel() for _, _, t, _ in entries], world)
 
     if mode == "shard":
@@ -117,9 +154,11 @@ def materialize_module_distributed(
                 submodule._parameters[key] = mat
             else:
                 submodule._buffers[key] = mat
+        _relink_duplicates(dups, first)
         return {i: owners[i] for i in range(len(entries))}
 
     _broadcast_bucketed(entries, owners, group, rank)
+    _relink_duplicates(dups, first)
     return {i: owners[i] for i in range(len(entries))}
 
 
@@ -181,6 +220,10 @@ def _broadcast_bucketed(entries, owners, group, rank) -> None:
 
     # ---- pipeline ---------------------------------------------------------
     pending = []  # (handle, stream) to drain at the end
+    # Tied parameters (one fake object in several slots) must swap to ONE
+    # materialized tensor on every rank; owners get this from
+    # materialize_tensor's identity stability, receivers from this map.
+    seen: Dict[int, torch.Tensor] = {}
     for k, (_, _, b) in enumerate(order):
         owner, dtype, idxs = buckets[b]
         stream = streams[k % 2] if use_streams else None
@@ -216,10 +259,14 @@ def _broadcast_bucketed(entries, owners, group, rank) -> None:
                     handle.wait()
                 for i, off in zip(idxs, offsets):
                     submodule, key, tensor, is_param = entries[i]
-                    mat = torch.empty(
-                        tensor.shape, dtype=tensor.dtype, device=tensor.device
-                    )
-                    mat.view(-1).copy_(flat[off : off + tensor.numel()])
+                    mat = seen.get(id(tensor))
+                    if mat is None:
+                        mat = torch.empty(
+                            tensor.shape, dtype=tensor.dtype,
+                            device=tensor.device,
+                        )
+                        mat.view(-1).copy_(flat[off : off + tensor.numel()])
+                        seen[id(tensor)] = mat
                     _swap_entry(entries[i], mat)
                 handle = None
         if handle is not None:
