@@ -351,6 +351,7 @@ def _materialize_allgather(
     entries = _named_deferred_tensors(module, buffers_only)
     if check_fn is not None:
         entries = [e for e in entries if check_fn(e[0])]
+    entries, dups, first = _dedupe_entries(entries)
 
     handles = []
     for submodule, key, tensor, is_param in entries:
@@ -383,6 +384,7 @@ def _materialize_allgather(
 
     for h in handles:
         h.wait()
+    _relink_duplicates(dups, first)
     return {}
 
 
