@@ -534,3 +534,80 @@ def test_batched_replay_preserves_tied_parameters() -> None:
     torch.cuda.synchronize()
     assert m.a.weight is m.b.weight
     assert m.a.weight.is_cuda and m.a.weight.shape == (32, 16)
+
+
+@pytest.mark.parametrize("seed", range(20))
+def test_batched_replay_fuzz(seed) -> None:
+    # Random models mixing plannable chains (uniform/normal/bernoulli/
+    # fill/zeros/ones/empty) with fallback-only tapes (trunc_normal
+    # pointwise tails, views, cross-tensor dependencies) and mixed
+    # dtypes: the batched planner must match per-tensor replay bitwise
+    # on every tensor.
+    import random
+
+    from torchdistx_amd import deferred_init
+    from torchdistx_amd.deferred_init import (
+        materialize_module,
+        materialize_module_batched,
+    )
+
+    rng = random.Random(seed + 3000)
+
+    def build():
+        class M(torch.nn.Module):
+            def __init__(self):
+                super().__init__()
+                for i in range(rng.randint(3, 10)):
+                    shape = rng.choice(
+                        [(7,), (64,), (33, 9), (128, 130), (5, 3, 17)]
+                    )
+                    dtype = rng.choice(
+                        [torch.float32, torch.bfloat16, torch.float16]
+                    )
+                    t = torch.empty(shape, device="cuda", dtype=dtype)
+                    kind = rng.choice(
+                        ["uniform", "normal", "bern", "fill", "zero",
+                         "ones", "empty0", "trunc", "dep"]
+                    )
+                    if kind == "uniform":
+                        t.uniform_(-1, 2)
+                    elif kind == "normal":
+                        t.normal_(0.1, 0.8)
+                    elif kind == "bern":
+                        t.bernoulli_(0.4)
+                    elif kind == "fill":
+                        t.fill_(2.5)
+                    elif kind == "zero":
+                        t.zero_()
+                    elif kind == "ones":
+                        t = torch.ones(shape, device="cuda", dtype=dtype)
+                    elif kind == "empty0":
+                        t.zero_()  # deterministic stand-in for empty
+                    elif kind == "trunc":
+                        torch.nn.init.trunc_normal_(t, 0.0, 0.5)
+                    else:  # dep: cross-tensor arithmetic -> fallback
+                        t = (
+                            torch.zeros(shape, device="cuda", dtype=dtype)
+                            + torch.ones(shape, device="cuda", dtype=dtype)
+                        )
+                    setattr(self, f"p{i}", torch.nn.Parameter(t))
+
+        return M()
+
+    rng_state = rng.getstate()
+    torch.manual_seed(seed)
+    rng.setstate(rng_state)
+    ref = deferred_init(build)
+    materialize_module(ref)
+
+    torch.manual_seed(seed)
+    rng.setstate(rng_state)
+    bat = deferred_init(build)
+    materialize_module_batched(bat)
+    torch.cuda.synchronize()
+
+    for (n1, p1), (n2, p2) in zip(
+        ref.named_parameters(), bat.named_parameters()
+    ):
+        assert n1 == n2
+        assert torch.equal(p1, p2), (seed, n1)
