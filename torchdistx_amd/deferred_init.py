@@ -226,8 +226,14 @@ def _batched_fill(entries) -> list:
             batch["entry"].append((entry, seen[id(tensor)]))
             continue
         plan = _C.tensor_init_plan(tensor)
-        if plan is None or plan["dtype"] not in (
-            torch.float32, torch.bfloat16, torch.float16
+        if (
+            plan is None
+            or plan["dtype"] not in (
+                torch.float32, torch.bfloat16, torch.float16
+            )
+            # One batched launch targets one device; stray tensors of a
+            # mixed-device module replay normally.
+            or plan["device"] != entries[0][2].device
         ):
             fallback.append(entry)
             continue
