@@ -168,6 +168,9 @@ def materialize_module_distributed(
 # starts while almost all init kernels are still pending.
 _BUCKET_BYTES = 128 << 20
 
+# Double-buffered side streams for the bucket pipeline (created once).
+_bucket_streams: List["torch.cuda.Stream"] = []
+
 
 def _broadcast_bucketed(entries, owners, group, rank) -> None:
     """Owner-materialize + broadcast, pipelined: entries are packed into
@@ -213,9 +216,15 @@ def _broadcast_bucketed(entries, owners, group, rank) -> None:
         and entries
         and entries[0][2].is_cuda
     )
-    streams = (
-        [torch.cuda.Stream(), torch.cuda.Stream()] if use_streams else None
-    )
+    # Stable process-lifetime streams: the caching allocator tags blocks
+    # with their allocation stream, so fresh Stream objects per call
+    # would orphan every cached bucket buffer and churn hipMalloc/Free
+    # on each step (measured ~90x slowdown for the analogous mistake in
+    # materialize_module_parallel).
+    global _bucket_streams
+    if use_streams and not _bucket_streams:
+        _bucket_streams = [torch.cuda.Stream(), torch.cuda.Stream()]
+    streams = _bucket_streams if use_streams else None
     from contextlib import nullcontext
 
     # ---- pipeline ---------------------------------------------------------
