@@ -31,3 +31,20 @@ def test_describe_module_counts() -> None:
     d2 = describe_module(m)
     assert d2["n_awaiting_materialization"] == 0
     assert d2["n_recorded_tensors"] == 0  # records dropped with the fakes
+
+
+def test_materialization_report() -> None:
+    from torchdistx_amd.utils.tape import materialization_report
+
+    m = deferred_init(Linear, 64, 64)
+    with materialization_report(m) as rep:
+        materialize_module(m)
+    assert rep["materialized_tensors"] == 2
+    # 64*64*4 + 64*4 bytes
+    assert rep["materialized_bytes"] == 64 * 64 * 4 + 64 * 4
+    assert rep["wall_s"] > 0 and rep["gbps"] > 0
+
+    # A second report on an already-real module measures nothing.
+    with materialization_report(m) as rep2:
+        pass
+    assert rep2["materialized_tensors"] == 0
