@@ -1,1 +1,5 @@
-from torchdistx_amd.utils.tape import describe_module, record_info  # noqa: F401
+from torchdistx_amd.utils.tape import (  # noqa: F401
+    describe_module,
+    materialization_report,
+    record_info,
+)
