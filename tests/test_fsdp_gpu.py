@@ -196,3 +196,30 @@ def test_shard_mode_batched_fill_bitwise(nccl_world1) -> None:
         ref.named_parameters(), m.named_parameters()
     ):
         assert n1 == n2 and torch.equal(p1, p2), n1
+
+
+def test_broadcast_and_allgather_modes_world1(nccl_world1) -> None:
+    # World-1 exercises the full bucketed-broadcast machinery (bucket
+    # planning, side-stream pipeline, RCCL broadcast-to-self, unpack
+    # bypass on the owner) and the allgather mode on real RCCL; results
+    # must match plain materialization bitwise.
+    from torchdistx_amd import deferred_init
+    from torchdistx_amd.deferred_init import materialize_module
+    from torchdistx_amd.models import TINY, build_model
+    from torchdistx_amd.parallel import materialize_module_distributed
+
+    for mode in ("broadcast", "allgather"):
+        torch.manual_seed(23)
+        ref = deferred_init(build_model, TINY, device="cuda",
+                            dtype=torch.bfloat16)
+        materialize_module(ref)
+
+        torch.manual_seed(23)
+        m = deferred_init(build_model, TINY, device="cuda",
+                          dtype=torch.bfloat16)
+        materialize_module_distributed(m, mode=mode)
+        torch.cuda.synchronize()
+        for (n1, p1), (n2, p2) in zip(
+            ref.named_parameters(), m.named_parameters()
+        ):
+            assert n1 == n2 and torch.equal(p1, p2), (mode, n1)
