@@ -476,3 +476,53 @@ def _tied_shard_worker(rank, world_size):
         assert not _C.can_materialize(m.a.weight)
     else:
         assert _C.can_materialize(m.a.weight)
+
+
+def test_broadcast_bucket_boundaries_multiproc() -> None:
+    # Many tensors, two dtypes, and a tiny bucket budget: forces multiple
+    # buckets per (owner, dtype) and exercises the pack/unpack offsets
+    # across bucket splits at world 2.
+    from tests._dist_utils import run_distributed
+
+    results = run_distributed(_bucket_boundary_worker, world_size=2)
+    assert results[0] == results[1]  # digests agree across ranks
+
+
+def _bucket_boundary_worker(rank, world_size):
+    import hashlib
+
+    import torch
+    from torch.nn import Module, Parameter
+
+    import torchdistx_amd.parallel.sharded_materialize as sm
+    from torchdistx_amd import _C, deferred_init
+    from torchdistx_amd.parallel import materialize_module_distributed
+
+    class M(Module):
+        def __init__(self):
+            super().__init__()
+            for i in range(13):
+                dt = torch.float32 if i % 2 else torch.bfloat16
+                n = [3, 17, 64, 129, 1000][i % 5]
+                self.register_parameter(
+                    f"p{i}", Parameter(torch.empty(n, dtype=dt).normal_())
+                )
+
+    saved = sm._BUCKET_BYTES
+    sm._BUCKET_BYTES = 512  # force many small buckets
+    try:
+        torch.manual_seed(777 + rank)  # skewed: the wire must reconcile
+        m = deferred_init(M)
+        _C.set_native_init_cpu(True)
+        try:
+            materialize_module_distributed(m, mode="broadcast")
+        finally:
+            _C.set_native_init_cpu(False)
+    finally:
+        sm._BUCKET_BYTES = saved
+
+    h = hashlib.sha256()
+    for name, p in sorted(m.named_parameters()):
+        h.update(name.encode())
+        h.update(p.detach().view(torch.uint8).numpy().tobytes())
+    return h.hexdigest()
