@@ -163,6 +163,7 @@ def materialize_module_batched(
     module: Module,
     buffers_only: bool = False,
     check_fn: Optional[Callable[[Module], bool]] = None,
+    device=None,
 ) -> None:
     """Materializes ``module`` through the batched replay planner: every
     parameter/buffer whose tape is a simple init chain collapses to its
@@ -172,7 +173,14 @@ def materialize_module_batched(
     (views, cross-tensor dependencies, pointwise tails) fall back to
     ordinary replay. GPU-only; CPU targets use :func:`materialize_module`.
 
-    Launch count for a Llama-3-70B replica drops from ~560 to ~3."""
+    Launch count for a Llama-3-70B replica drops from ~560 to ~3.
+
+    ``device`` retargets materialization: plans allocate and fill
+    directly on that device (e.g. a tape recorded with device="cpu"
+    lands straight in HBM; the pinned counters make uniform/bernoulli/
+    fill bits identical across devices, normals equivalent but
+    per-device-transformed). Fallback tensors replay on their recorded
+    device and are then moved."""
     import torch
 
     from torchdistx_amd import _kernels
@@ -189,27 +197,40 @@ def materialize_module_batched(
             if b is not None and _C.can_materialize(b):
                 entries.append((submodule, key, b, False))
 
-    for entry in _batched_fill(entries):
+    for entry in _batched_fill(entries, device=device):
         submodule, key, tensor, is_param = entry
         mat = materialize_tensor(tensor)
+        if device is not None and mat.device != torch.device(device):
+            moved = mat.detach().to(device)
+            moved.requires_grad_(mat.requires_grad)
+            mat = _restore_class(tensor, moved)
         if is_param:
             submodule._parameters[key] = mat
         else:
             submodule._buffers[key] = mat
 
 
-def _batched_fill(entries) -> list:
+def _batched_fill(entries, device=None) -> list:
     """Fills every plannable (simple-chain, GPU, f32/bf16/f16) entry with
     one batched kernel launch and swaps the module slots; returns the
     entries that must go through ordinary replay instead. Used by
-    :func:`materialize_module_batched` and the distributed shard path."""
+    :func:`materialize_module_batched` and the distributed shard path.
+    ``device`` overrides the plans' target device (retargeting)."""
     import torch
 
     from torchdistx_amd import _kernels
 
+    if device is not None:
+        device = torch.device(device)
+    target_cuda = (
+        device.type == "cuda"
+        if device is not None
+        else bool(entries) and entries[0][2].is_cuda
+    )
     gpu = (
         bool(entries)
-        and entries[0][2].is_cuda
+        and target_cuda
+        and torch.cuda.is_available()
         and getattr(_kernels, "_K", None) is not None
         and _kernels._K.has_batched_init()
     )
@@ -233,12 +254,14 @@ def _batched_fill(entries) -> list:
             )
             # One batched launch targets one device; stray tensors of a
             # mixed-device module replay normally.
-            or plan["device"] != entries[0][2].device
+            or (device is None and plan["device"] != entries[0][2].device)
         ):
             fallback.append(entry)
             continue
         out = torch.empty(
-            plan["sizes"], dtype=plan["dtype"], device=plan["device"]
+            plan["sizes"],
+            dtype=plan["dtype"],
+            device=device if device is not None else plan["device"],
         )
         seen[id(tensor)] = out
         code = _DIST_CODE[plan["kind"]]
