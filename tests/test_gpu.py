@@ -643,7 +643,11 @@ def test_retarget_cpu_tape_to_gpu() -> None:
 
     torch.manual_seed(71)
     m = deferred_init(M)
-    materialize_module_batched(m, device="cuda")
+    _C.set_native_init_cpu(True)  # fallback chain replays pinned on CPU
+    try:
+        materialize_module_batched(m, device="cuda")
+    finally:
+        _C.set_native_init_cpu(False)
     torch.cuda.synchronize()
     assert all(p.is_cuda for p in m.parameters())
     assert torch.equal(m.u.cpu(), ref.u.detach())
