@@ -656,3 +656,23 @@ def test_materialize_module_batched_cpu_fallback() -> None:
     torch.manual_seed(88)
     e = torch.nn.Linear(8, 8)
     assert torch.equal(m.weight, e.weight)
+
+
+def test_rng_on_noncontiguous_view_matches_eager() -> None:
+    # The in-place record fast path must compose with views: normal_ on
+    # a transposed (non-contiguous) view replays through stock ATen
+    # (the native kernels require contiguity and decline) and must
+    # reproduce eager bits in tape order.
+    class M(Module):
+        def __init__(self):
+            super().__init__()
+            a = torch.zeros(4, 6)
+            a.t().normal_(0.0, 1.0)
+            self.p = Parameter(a)
+
+    torch.manual_seed(321)
+    m = deferred_init(M)
+    materialize_module(m)
+    torch.manual_seed(321)
+    e = M()
+    assert torch.equal(m.p, e.p)
