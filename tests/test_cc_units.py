@@ -1,0 +1,24 @@
+# Builds and runs the standalone C++ unit tests (tests/cc/) with plain
+# g++ — covering the header-only pieces (Philox pipeline) both the CPU
+# impls and the CDNA4 kernels compile. The reference planned C++ tests
+# but never added them (reference CMakeLists.txt:104-106).
+
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_philox_cc_units(tmp_path) -> None:
+    binary = tmp_path / "test_philox"
+    build = subprocess.run(
+        ["g++", "-O2", "-std=c++17", os.path.join(REPO, "tests", "cc",
+         "test_philox.cc"), "-o", str(binary)],
+        capture_output=True, text=True, timeout=120,
+    )
+    assert build.returncode == 0, build.stderr
+    run = subprocess.run([str(binary)], capture_output=True, text=True,
+                         timeout=120)
+    assert run.returncode == 0, run.stdout + run.stderr
+    assert "all passed" in run.stdout
