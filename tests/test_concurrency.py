@@ -190,3 +190,89 @@ def test_concurrent_sessions_stress() -> None:
             t.join()
         assert not errors, errors
         assert not torch.equal(results[0].weight, results[1].weight), trial
+
+
+def test_chaos_mixed_tape_operations() -> None:
+    # Four threads hammer one process with overlapping tape operations:
+    # materialization, slice materialization, introspection, and fresh
+    # recording sessions — probing lock interleavings between the
+    # structure lock, unlocked op execution, the RNG registry, and the
+    # hooks proxy. Any torn state surfaces as wrong values or a crash.
+    import random
+
+    from torchdistx_amd import _C, deferred_init, materialize_tensor
+    from torchdistx_amd.utils import describe_module, record_info
+
+    def build(n):
+        return nn.Sequential(*[nn.Linear(24, 24) for _ in range(n)])
+
+    errors = []
+    torch.manual_seed(4321)
+    shared = deferred_init(build, 6)
+    expected = {}
+    _C.set_native_init_cpu(True)
+    try:
+        torch.manual_seed(4321)
+        ref = deferred_init(build, 6)
+        for name, p in ref.named_parameters():
+            expected[name] = _C.materialize_tensor(p).detach().clone()
+    finally:
+        _C.set_native_init_cpu(False)
+
+    barrier = threading.Barrier(4)
+
+    def materializer():
+        try:
+            barrier.wait()
+            _C.set_native_init_cpu(True)
+            params = list(shared.named_parameters())
+            random.Random(1).shuffle(params)
+            for name, p in params:
+                out = materialize_tensor(p)
+                assert torch.equal(out.detach(), expected[name]), name
+        except Exception as e:
+            errors.append(e)
+
+    def slicer():
+        try:
+            barrier.wait()
+            for _ in range(30):
+                m = deferred_init(build, 2)
+                w = m[0].weight
+                try:
+                    shard = _C.materialize_tensor_shard(w, 3, 9)
+                    assert shard.shape == (6, 24)
+                except RuntimeError:
+                    pass  # chain freed by a racing materialize: legal
+        except Exception as e:
+            errors.append(e)
+
+    def introspector():
+        try:
+            barrier.wait()
+            for _ in range(60):
+                describe_module(shared)
+                for p in shared.parameters():
+                    record_info(p)
+        except Exception as e:
+            errors.append(e)
+
+    def recorder():
+        try:
+            barrier.wait()
+            for i in range(25):
+                m = deferred_init(build, 3)
+                materialize_tensor(m[i % 3].weight)
+        except Exception as e:
+            errors.append(e)
+
+    threads = [
+        threading.Thread(target=f)
+        for f in (materializer, slicer, introspector, recorder)
+    ]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    _C.set_native_init_cpu(False)
+    assert not errors, errors
