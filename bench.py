@@ -36,7 +36,8 @@ def get_args():
     p.add_argument("--steps", type=int, default=3)
     p.add_argument("--warmup", type=int, default=1)
     p.add_argument("--model", type=str, default=None,
-                   help="config name (default: llama3-70b on GPU, tiny on CPU)")
+                   help="config name (default: llama3-70b on GPU, tiny on "
+                        "CPU); see torchdistx_amd.models.CONFIGS")
     p.add_argument("--mode", type=str, default="replicate",
                    choices=["replicate", "shard", "broadcast", "allgather", "slice"])
     p.add_argument("--dtype", type=str, default="bf16",
@@ -113,6 +114,10 @@ def main():
         sys.exit(run_selftest(args, dist, device, rank, world, use_cuda))
 
     model_name = args.model or ("llama3-70b" if use_cuda else "tiny")
+    if model_name not in CONFIGS:
+        print(f"FATAL: unknown model {model_name!r}; available: "
+              f"{sorted(CONFIGS)}", file=sys.stderr)
+        sys.exit(2)
     cfg = CONFIGS[model_name]
     if args.init != cfg.init:
         import dataclasses
