@@ -653,3 +653,25 @@ def test_retarget_cpu_tape_to_gpu() -> None:
     assert torch.equal(m.u.cpu(), ref.u.detach())
     assert torch.equal(m.f.cpu(), ref.f.detach())
     assert torch.equal(m.t.cpu(), ref.t.detach())  # replayed on CPU, moved
+
+
+def test_expert_sharded_materialization_gpu() -> None:
+    # Expert parallelism at init time on GPU: only this rank's experts
+    # materialize, shared params everywhere, no communication (rank 0 of
+    # a simulated world of 1 owns every expert; the multiproc ownership
+    # math is covered at world 2/4 over gloo in tests/test_parallel.py).
+    from torchdistx_amd import deferred_init, is_deferred
+    from torchdistx_amd.models import CONFIGS, build_model
+    from torchdistx_amd.parallel import materialize_experts_sharded
+
+    cfg = CONFIGS["tiny-moe"]
+    torch.manual_seed(13)
+    m = deferred_init(build_model, cfg, device="cuda", dtype=torch.bfloat16)
+    owners = materialize_experts_sharded(m)
+    assert owners and all(o == 0 for o in owners.values())
+    assert not is_deferred(m)
+    tokens = torch.randint(0, cfg.vocab_size, (2, 16), device="cuda")
+    loss = m.loss(tokens)
+    loss.backward()
+    torch.cuda.synchronize()
+    assert loss.isfinite().item()
