@@ -39,9 +39,10 @@ namespace {
 constexpr int kBlock = 256;
 // Measured on gfx950 (scripts/rng_tune.hip): this ALU-heavy streaming
 // kernel keeps improving past the usual ~2048-block guideline — at 16384
-// blocks the one-philox-per-store uniform kernel reaches 5.55 TB/s on a
-// 4 GiB bf16 fill (pure-store ceiling: 6.37 TB/s via hipMemset; normals
-// are transcendental-bound at ~2.7 TB/s).
+// blocks the one-philox-per-store uniform kernel reaches ~5.5 TB/s on a
+// 4 GiB bf16 fill (pure-store ceiling: 6.4 TB/s via hipMemset; the
+// Philox-7 + log2-Box-Muller normals are VALU-bound at ~3.5 TB/s once
+// clocks are warm — profiles/dvfs_ramp_note.md).
 constexpr int kMaxBlocks = 16384;
 
 // ---------------------------------------------------------------------------
