@@ -38,7 +38,17 @@ __device__ __forceinline__ float loadf(const void* p, DT dt, int64_t i) {
   }
 }
 
+// Pins the f32 rounding step before narrowing conversions, so every
+// code path (generic, vectorized, batched) rounds identically — the
+// compiler otherwise fuses fma+convert (v_fma_mix*) in some shapes only
+// (see init_kernels.hip from_float).
+__device__ __forceinline__ float roundBarrier(float v) {
+  asm volatile("" : "+v"(v));
+  return v;
+}
+
 __device__ __forceinline__ void storef(void* p, DT dt, int64_t i, float v) {
+  v = roundBarrier(v);
   switch (dt) {
     case DT::kF32:
       static_cast<float*>(p)[i] = v;
@@ -79,12 +89,13 @@ __device__ __forceinline__ void store8(T* p, int64_t base, const float* in) {
   Pack8Aligned<T> pk;
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
+    const float vj = roundBarrier(in[j]);
     if constexpr (std::is_same_v<T, float>) {
-      pk.v[j] = in[j];
+      pk.v[j] = vj;
     } else if constexpr (std::is_same_v<T, __hip_bfloat16>) {
-      pk.v[j] = __float2bfloat16(in[j]);
+      pk.v[j] = __float2bfloat16(vj);
     } else {
-      pk.v[j] = __float2half(in[j]);
+      pk.v[j] = __float2half(vj);
     }
   }
   *reinterpret_cast<Pack8Aligned<T>*>(p + base) = pk;
@@ -142,12 +153,13 @@ __global__ void adamw_vec_kernel(Tp* __restrict__ param,
         // Round-trip through the storage dtype to keep the Kahan
         // remainder exact w.r.t. what is actually stored.
         Tp stored;
+        const float p_new_r = roundBarrier(p_new);
         if constexpr (std::is_same_v<Tp, float>) {
-          stored = p_new;
+          stored = p_new_r;
         } else if constexpr (std::is_same_v<Tp, __hip_bfloat16>) {
-          stored = __float2bfloat16(p_new);
+          stored = __float2bfloat16(p_new_r);
         } else {
-          stored = __float2half(p_new);
+          stored = __float2half(p_new_r);
         }
         float p_rounded;
         if constexpr (std::is_same_v<Tp, float>) {

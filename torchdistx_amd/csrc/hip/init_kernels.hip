@@ -118,6 +118,15 @@ __device__ __forceinline__ float2 box_muller(float u1, float u2) {
 
 template <typename T>
 __device__ __forceinline__ T from_float(float v) {
+  // Pin the f32 rounding step: without this barrier the compiler fuses
+  // fma+convert into v_fma_mixlo_f16 (ONE rounding, straight to fp16)
+  // in some loop shapes but not others, so the same element's bits
+  // depended on which code path (vectorized main loop vs elementwise
+  // boundary) stored it — a 1-ulp fp16 divergence that broke the
+  // slice/full bitwise invariant (caught by the 40k-seed GPU slice
+  // fuzz). bf16/f32 paths never fused; the barrier costs one cvt that
+  // was already in the unfused schedule.
+  asm volatile("" : "+v"(v));
   if constexpr (std::is_same_v<T, __hip_bfloat16>) {
     return __float2bfloat16(v);
   } else if constexpr (std::is_same_v<T, __half>) {
