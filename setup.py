@@ -87,7 +87,7 @@ setup(
         "torchdistx_amd.ops",
         "torchdistx_amd.utils",
     ],
-    package_data={"torchdistx_amd": ["_C.pyi", "py.typed"]},
+    package_data={"torchdistx_amd": ["_C.pyi", "_K.pyi", "py.typed"]},
     ext_modules=ext_modules,
     cmdclass={"build_ext": cpp_extension.BuildExtension},
 )
