@@ -526,3 +526,20 @@ def _bucket_boundary_worker(rank, world_size):
         h.update(name.encode())
         h.update(p.detach().view(torch.uint8).numpy().tobytes())
     return h.hexdigest()
+
+
+def test_broadcast_pipeline_fuzz() -> None:
+    # Bounded slice of scripts/broadcast_fuzz.py: random models (counts,
+    # shapes, dtypes, tied params), random bucket budgets, rank-skewed
+    # seeds half the time, worlds 2-4 — every rank must converge to one
+    # digest through the bucketed wire.
+    import os
+    import subprocess
+    import sys as _sys
+
+    result = subprocess.run(
+        [_sys.executable, "scripts/broadcast_fuzz.py", "6", "100"],
+        capture_output=True, text=True, timeout=500,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+    )
+    assert result.returncode == 0, result.stdout[-1500:] + result.stderr[-500:]
