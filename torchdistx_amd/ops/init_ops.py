@@ -72,3 +72,20 @@ def zero_(tensor: torch.Tensor) -> torch.Tensor:
     if _native(tensor):
         return torch.ops.tdx.zero_(tensor)
     return tensor.zero_()
+
+
+def bernoulli_(
+    tensor: torch.Tensor,
+    p: float = 0.5,
+    *,
+    seed: Optional[int] = None,
+    offset: Optional[int] = None,
+) -> torch.Tensor:
+    """Bernoulli(p) 0/1 fill (uniform-threshold over the Philox stream)."""
+    if _native(tensor) and tensor.dtype in _RNG_DTYPES:
+        return torch.ops.tdx.bernoulli_(tensor, p, seed=seed, offset=offset)
+    if seed is not None:
+        g = torch.Generator(device=tensor.device)
+        g.manual_seed(seed + (offset or 0))
+        return tensor.bernoulli_(p, generator=g)
+    return tensor.bernoulli_(p)
