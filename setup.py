@@ -18,7 +18,7 @@ os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 ROOT = os.path.dirname(os.path.abspath(__file__))
 CSRC = os.path.join(ROOT, "torchdistx_amd", "csrc")
 
-BASE_VERSION = "0.1.0"
+BASE_VERSION = "0.2.0"
 
 
 def version_with_variant() -> str:

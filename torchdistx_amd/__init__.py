@@ -10,13 +10,15 @@
 #   materialize_tensor / materialize_module — reference deferred_init.py:19-124
 #   slowmo, gossip_grad, optimizers        — reference slowmo/, gossip_grad.py, optimizers/
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 from torchdistx_amd.fake import fake_mode, is_fake, meta_like  # noqa: F401
 from torchdistx_amd.deferred_init import (  # noqa: F401
     deferred_init,
     is_deferred,
     materialize_module,
+    materialize_module_batched,
+    materialize_module_parallel,
     materialize_tensor,
 )
 
