@@ -11,6 +11,13 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
 def test_philox_cc_units(tmp_path) -> None:
+    import pytest
+
+    header = os.path.join(
+        REPO, "torchdistx_amd", "csrc", "core", "philox.h"
+    )
+    if not os.path.exists(header):
+        pytest.skip("C++ sources not present (wheel-installed run)")
     binary = tmp_path / "test_philox"
     build = subprocess.run(
         ["g++", "-O2", "-std=c++17", os.path.join(REPO, "tests", "cc",

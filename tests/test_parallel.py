@@ -537,9 +537,12 @@ def test_broadcast_pipeline_fuzz() -> None:
     import subprocess
     import sys as _sys
 
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    if not os.path.exists(os.path.join(repo, "scripts", "broadcast_fuzz.py")):
+        pytest.skip("scripts/ not present (wheel-installed run)")
     result = subprocess.run(
         [_sys.executable, "scripts/broadcast_fuzz.py", "6", "100"],
         capture_output=True, text=True, timeout=500,
-        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+        cwd=repo,
     )
     assert result.returncode == 0, result.stdout[-1500:] + result.stderr[-500:]
