@@ -39,6 +39,19 @@ def main():
 
     cfg = LLAMA3_8B if os.environ.get("SMALL") else LLAMA3_405B
 
+    # Fail fast with a clear message instead of an allocator OOM when the
+    # per-rank slice cannot fit (e.g. the full 812 GB model at world 1).
+    per_rank = -(-cfg.n_params * 2 // world)  # bf16 bytes, ceil
+    hbm = torch.cuda.get_device_properties(0).total_memory
+    if per_rank > hbm:
+        if rank == 0:
+            print(f"FATAL: {cfg.name} needs {per_rank / 1e9:.0f} GB per "
+                  f"rank at world {world} but the device has "
+                  f"{hbm / 1e9:.0f} GB; run with more ranks, or SMALL=1 "
+                  f"for the Llama-3-8B variant.", file=sys.stderr)
+        dist.destroy_process_group()
+        sys.exit(1)
+
     torch.manual_seed(0)  # all ranks pin the same Philox streams
     t0 = time.perf_counter()
     model = deferred_init(build_model, cfg, device="cuda",
