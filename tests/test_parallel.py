@@ -348,6 +348,149 @@ def test_slice_materialization_pointwise_chain() -> None:
         _C.set_native_init_cpu(False)
 
 
+def test_slice_materialization_any_dim_matches_full() -> None:
+    # Any-dim slices (the windowed shard path, n_blocks > 1) are bitwise
+    # sub-tensors of the full native materialization — group-aligned and
+    # odd geometries, every dtype, every init kind.
+    from torch.nn import Module, Parameter
+
+    from torchdistx_amd import _C, deferred_init
+
+    cases = [
+        # (shape, dim, ranges) — (24, 40) is 8-aligned in the trailing
+        # dim; (9, 11, 7) is odd everywhere (elementwise fallback).
+        ((24, 40), 1, [(0, 40), (0, 8), (16, 40), (13, 27), (5, 5)]),
+        ((9, 11, 7), 1, [(0, 11), (3, 8), (10, 11)]),
+        ((9, 11, 7), 2, [(0, 7), (2, 5)]),
+        ((6, 8, 16), 2, [(0, 16), (8, 16)]),
+    ]
+    for dtype in (torch.float32, torch.bfloat16, torch.float16):
+        for init in ("normal", "uniform", "bernoulli", "trunc"):
+            for shape, dim, ranges in cases:
+
+                class M(Module):
+                    def __init__(self):
+                        super().__init__()
+                        w = torch.empty(shape, dtype=dtype)
+                        if init == "normal":
+                            w.normal_(0.2, 1.3)
+                        elif init == "uniform":
+                            w.uniform_(-2.0, 3.0)
+                        elif init == "bernoulli":
+                            w.bernoulli_(0.4)
+                        else:
+                            torch.nn.init.trunc_normal_(w, std=0.5)
+                        self.p = Parameter(w)
+
+                _C.set_native_init_cpu(True)
+                try:
+                    torch.manual_seed(777)
+                    full = _C.materialize_tensor(deferred_init(M).p).detach()
+                    torch.manual_seed(777)
+                    part = deferred_init(M)
+                    for a, b in ranges:
+                        shard = _C.materialize_tensor_shard(part.p, a, b, dim)
+                        assert torch.equal(
+                            shard, full.narrow(dim, a, b - a)
+                        ), (dtype, init, shape, dim, a, b)
+                finally:
+                    _C.set_native_init_cpu(False)
+
+
+def test_dim1_slices_reassemble_and_validate() -> None:
+    # Concatenating every rank's dim-1 slice reconstructs the exact full
+    # tensor (the row-parallel TP contract); bogus dims/ranges raise.
+    import pytest
+
+    from torch.nn import Module, Parameter
+
+    from torchdistx_amd import _C, deferred_init
+    from torchdistx_amd.parallel import materialize_tensor_shard
+
+    class M(Module):
+        def __init__(self):
+            super().__init__()
+            self.p = Parameter(torch.empty(16, 48).normal_())
+
+    _C.set_native_init_cpu(True)
+    try:
+        torch.manual_seed(88)
+        full = _C.materialize_tensor(deferred_init(M).p).detach()
+        torch.manual_seed(88)
+        part = deferred_init(M)
+        world = 3
+        cols = [
+            materialize_tensor_shard(
+                part.p, r * 48 // world, (r + 1) * 48 // world, dim=1
+            )
+            for r in range(world)
+        ]
+        assert torch.equal(torch.cat([c.detach() for c in cols], dim=1), full)
+        assert cols[0].requires_grad
+
+        with pytest.raises(RuntimeError, match="invalid slice dim"):
+            _C.materialize_tensor_shard(part.p, 0, 4, 2)
+        with pytest.raises(RuntimeError, match="invalid slice range"):
+            _C.materialize_tensor_shard(part.p, 0, 49, 1)
+    finally:
+        _C.set_native_init_cpu(False)
+
+
+def test_tp_sharded_module() -> None:
+    # materialize_module_tp_sharded: listed tensors shard along their
+    # assigned dim, unlisted ones replicate bitwise, unknown names raise.
+    import pytest
+
+    from torch.nn import Linear, Module
+
+    from torchdistx_amd import _C, deferred_init
+    from torchdistx_amd.parallel import materialize_module_tp_sharded
+
+    class Block(Module):
+        def __init__(self):
+            super().__init__()
+            self.up = Linear(32, 64)    # column-parallel: shard dim 0
+            self.down = Linear(64, 32)  # row-parallel: shard dim 1
+
+    shard_dims = {"up.weight": 0, "up.bias": 0, "down.weight": 1}
+
+    _C.set_native_init_cpu(True)
+    try:
+        torch.manual_seed(99)
+        full = deferred_init(Block)
+        from torchdistx_amd.deferred_init import materialize_module
+
+        materialize_module(full)
+        reference = dict(full.named_parameters())
+
+        world = 2
+        gathered: dict = {}
+        for rank in range(world):
+            torch.manual_seed(99)
+            m = deferred_init(Block)
+            out = materialize_module_tp_sharded(
+                m, shard_dims, rank=rank, world_size=world
+            )
+            for name, t in out.items():
+                gathered.setdefault(name, []).append(t.detach())
+
+        for name, parts in gathered.items():
+            ref = reference[name].detach()
+            dim = shard_dims.get(name)
+            if dim is None:  # replicated (down.bias): identical full copies
+                assert torch.equal(parts[0], parts[1]), name
+                assert torch.equal(parts[0], ref), name
+            else:
+                assert torch.equal(torch.cat(parts, dim=dim), ref), name
+
+        torch.manual_seed(99)
+        m = deferred_init(Block)
+        with pytest.raises(ValueError, match="not found in the module"):
+            materialize_module_tp_sharded(m, {"nope.weight": 0})
+    finally:
+        _C.set_native_init_cpu(False)
+
+
 def test_bernoulli_native_and_slice() -> None:
     # bernoulli_ joins the pinned-Philox op set: native CPU replay draws
     # from the pinned counters and the shard path reproduces any row

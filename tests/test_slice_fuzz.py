@@ -1,9 +1,11 @@
 # Differential fuzz of SLICE materialization boundary arithmetic: random
 # simple init chains (factory -> whole-tensor RNG/fill -> detach
-# passthroughs), random dim-0 row ranges, odd row lengths — the row-start
-# element offsets deliberately misalign with the kernels' 8-element Philox
-# groups. Every slice must be bitwise-equal to the same rows of a full
-# materialization of an INDEPENDENT tape recorded from the same seed.
+# passthroughs), random ranges along a RANDOM dim, odd lengths — the
+# slice-start element offsets deliberately misalign with the kernels'
+# 8-element Philox groups, and dim > 0 takes the windowed shard kernels
+# (n_blocks > 1). Every slice must be bitwise-equal to the same window of
+# a full materialization of an INDEPENDENT tape recorded from the same
+# seed.
 
 import random
 
@@ -93,18 +95,22 @@ def _run_case(seed: int, device: str) -> None:
     torch.manual_seed(seed)
     full = _C.materialize_tensor(deferred_init(build).t)
 
-    # ...random slices from fresh, independent tapes.
-    rows = shape[0]
+    # ...random slices along a random dim from fresh, independent tapes
+    # (dim > 0 exercises the windowed shard kernels).
     for _ in range(3):
-        start = rng.randint(0, rows)
-        end = rng.randint(start, rows)
+        dim = rng.randrange(len(shape))
+        n = shape[dim]
+        start = rng.randint(0, n)
+        end = rng.randint(start, n)
         rng2 = random.Random(seed + 1)
         torch.manual_seed(seed)
         holder = deferred_init(build)
-        shard = _C.materialize_tensor_shard(holder.t, start, end)
-        assert shard.shape == (end - start,) + tuple(shape[1:])
-        assert torch.equal(shard, full[start:end]), (
-            seed, shape, dtype, start, end
+        shard = _C.materialize_tensor_shard(holder.t, start, end, dim)
+        expect_shape = list(shape)
+        expect_shape[dim] = end - start
+        assert shard.shape == tuple(expect_shape)
+        assert torch.equal(shard, full.narrow(dim, start, end - start)), (
+            seed, shape, dtype, dim, start, end
         )
 
 

@@ -1262,8 +1262,20 @@ ChainStep classifyChainNode(const OpNode& node) {
   return step;
 }
 
-void applyShardStep(const ChainStep& step, at::Tensor& shard, int64_t start,
-                    int64_t end) {
+// A dim-d slice of a contiguous tensor, flattened: n_blocks contiguous
+// global element ranges of block_len elements, block r starting at
+// g_off + r * g_stride. dim 0 is the n_blocks == 1 case and takes the
+// flat shard ops (whose boundary handling is tuned for it); dim > 0
+// takes the windowed ops.
+struct ShardWindow {
+  int64_t n_blocks;
+  int64_t block_len;
+  int64_t g_stride;
+  int64_t g_off;
+};
+
+void applyShardStep(const ChainStep& step, at::Tensor& shard,
+                    const ShardWindow& w) {
   switch (step.kind) {
     case ChainStep::Kind::kFactory:
     case ChainStep::Kind::kPass:
@@ -1307,19 +1319,45 @@ void applyShardStep(const ChainStep& step, at::Tensor& shard, int64_t start,
               .findSchemaOrThrow("tdx::bernoulli_shard_", "")
               .typed<at::Tensor&(at::Tensor&, int64_t, int64_t, double,
                                  int64_t, int64_t)>();
+      static const auto uniform_shard_win =
+          c10::Dispatcher::singleton()
+              .findSchemaOrThrow("tdx::uniform_shard_win_", "")
+              .typed<at::Tensor&(at::Tensor&, int64_t, int64_t, int64_t,
+                                 int64_t, double, double, int64_t,
+                                 int64_t)>();
+      static const auto normal_shard_win =
+          c10::Dispatcher::singleton()
+              .findSchemaOrThrow("tdx::normal_shard_win_", "")
+              .typed<at::Tensor&(at::Tensor&, int64_t, int64_t, int64_t,
+                                 int64_t, double, double, int64_t,
+                                 int64_t)>();
+      static const auto bernoulli_shard_win =
+          c10::Dispatcher::singleton()
+              .findSchemaOrThrow("tdx::bernoulli_shard_win_", "")
+              .typed<at::Tensor&(at::Tensor&, int64_t, int64_t, int64_t,
+                                 int64_t, double, int64_t, int64_t)>();
       const auto& [seed, offset] = *step.philox;
-      if (step.kind == ChainStep::Kind::kUniform) {
-        uniform_shard.call(shard, start, end, step.p0, step.p1,
-                           static_cast<int64_t>(seed),
-                           static_cast<int64_t>(offset));
+      const auto s64 = static_cast<int64_t>(seed);
+      const auto o64 = static_cast<int64_t>(offset);
+      if (w.n_blocks == 1) {
+        const int64_t start = w.g_off;
+        const int64_t end = w.g_off + w.block_len;
+        if (step.kind == ChainStep::Kind::kUniform) {
+          uniform_shard.call(shard, start, end, step.p0, step.p1, s64, o64);
+        } else if (step.kind == ChainStep::Kind::kNormal) {
+          normal_shard.call(shard, start, end, step.p0, step.p1, s64, o64);
+        } else {
+          bernoulli_shard.call(shard, start, end, step.p0, s64, o64);
+        }
+      } else if (step.kind == ChainStep::Kind::kUniform) {
+        uniform_shard_win.call(shard, w.n_blocks, w.block_len, w.g_stride,
+                               w.g_off, step.p0, step.p1, s64, o64);
       } else if (step.kind == ChainStep::Kind::kNormal) {
-        normal_shard.call(shard, start, end, step.p0, step.p1,
-                          static_cast<int64_t>(seed),
-                          static_cast<int64_t>(offset));
+        normal_shard_win.call(shard, w.n_blocks, w.block_len, w.g_stride,
+                              w.g_off, step.p0, step.p1, s64, o64);
       } else {
-        bernoulli_shard.call(shard, start, end, step.p0,
-                             static_cast<int64_t>(seed),
-                             static_cast<int64_t>(offset));
+        bernoulli_shard_win.call(shard, w.n_blocks, w.block_len, w.g_stride,
+                                 w.g_off, step.p0, s64, o64);
       }
       return;
     }
@@ -1404,18 +1442,34 @@ std::optional<InitPlan> tensorInitPlan(const at::Tensor& tensor) {
 
 at::Tensor materializeTensorShard(const at::Tensor& tensor,
                                   int64_t start_row,
-                                  int64_t end_row) {
+                                  int64_t end_row,
+                                  int64_t dim) {
   auto* fake = asFake(tensor);
   TORCH_CHECK_VALUE(fake != nullptr && getRecord(fake) != nullptr,
                     "`tensor` is not a deferred tensor.");
   const at::Tensor& meta = fake->meta_tensor();
   TORCH_CHECK(meta.is_contiguous(),
               "slice materialization requires a contiguous tensor");
-  const int64_t rows = meta.dim() == 0 ? 1 : meta.size(0);
-  TORCH_CHECK(0 <= start_row && start_row <= end_row && end_row <= rows,
-              "invalid row range [", start_row, ", ", end_row, ") for ",
-              rows, " rows");
-  const int64_t row_elems = rows == 0 ? 0 : meta.numel() / rows;
+  TORCH_CHECK(0 <= dim && (dim < meta.dim() || (dim == 0 && meta.dim() == 0)),
+              "invalid slice dim ", dim, " for a ", meta.dim(),
+              "-d tensor");
+  const int64_t dim_len = meta.dim() == 0 ? 1 : meta.size(dim);
+  TORCH_CHECK(0 <= start_row && start_row <= end_row && end_row <= dim_len,
+              "invalid slice range [", start_row, ", ", end_row,
+              ") for size ", dim_len, " along dim ", dim);
+  // full.narrow(dim, start, end - start) flattened = a strided window
+  // over the full element space: prod(sizes[:dim]) blocks of
+  // (end - start) * prod(sizes[dim+1:]) contiguous elements each.
+  int64_t inner = 1;  // prod(sizes[dim+1:])
+  for (int64_t d = dim + 1; d < meta.dim(); ++d) {
+    inner *= meta.size(d);
+  }
+  int64_t n_blocks = 1;  // prod(sizes[:dim])
+  for (int64_t d = 0; d < dim; ++d) {
+    n_blocks *= meta.size(d);
+  }
+  const ShardWindow w{n_blocks, (end_row - start_row) * inner,
+                      dim_len * inner, start_row * inner};
 
   std::lock_guard<std::recursive_mutex> lock{tape_mutex};
   auto rec = getRecord(fake);
@@ -1434,16 +1488,14 @@ at::Tensor materializeTensorShard(const at::Tensor& tensor,
 
   std::vector<int64_t> shard_sizes(meta.sizes().begin(), meta.sizes().end());
   if (!shard_sizes.empty()) {
-    shard_sizes[0] = end_row - start_row;
+    shard_sizes[dim] = end_row - start_row;
   }
   at::Tensor shard = at::empty(
       shard_sizes, at::TensorOptions()
                        .dtype(meta.scalar_type())
                        .device(fake->fake_device()));
-  const int64_t start = start_row * row_elems;
-  const int64_t end = end_row * row_elems;
   for (const ChainStep& step : steps) {
-    applyShardStep(step, shard, start, end);
+    applyShardStep(step, shard, w);
   }
   return shard;
 }

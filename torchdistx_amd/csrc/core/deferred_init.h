@@ -73,9 +73,11 @@ struct InitPlan {
 
 std::optional<InitPlan> tensorInitPlan(const at::Tensor& tensor);
 
-// Slice materialization: materializes rows [start_row, end_row) of the
-// deferred tensor's dim 0 WITHOUT materializing the rest, bitwise-equal to
-// the corresponding slice of a full materialization (per device type).
+// Slice materialization: materializes indices [start_row, end_row) of
+// the deferred tensor along `dim` WITHOUT materializing the rest,
+// bitwise-equal to the corresponding slice of a full materialization
+// (per device type). dim 0 serves FSDP `Shard(0)` and column-parallel
+// weights; dim 1 serves row-parallel (Megatron-style TP) weights.
 // Requires the tensor's tape to be a "simple init chain" — a factory
 // (empty/zeros/ones/full) followed by whole-tensor in-place init ops
 // (uniform_/normal_/fill_/zero_) and aliasing pass-throughs
@@ -86,6 +88,7 @@ std::optional<InitPlan> tensorInitPlan(const at::Tensor& tensor);
 // with zero communication.
 at::Tensor materializeTensorShard(const at::Tensor& tensor,
                                   int64_t start_row,
-                                  int64_t end_row);
+                                  int64_t end_row,
+                                  int64_t dim = 0);
 
 }  // namespace tdx

@@ -135,15 +135,18 @@ PYBIND11_MODULE(_C, m) {
      "batched replay planner (materialize_module_batched).");
 
   m.def("materialize_tensor_shard",
-        [](const at::Tensor& t, int64_t start_row, int64_t end_row) {
+        [](const at::Tensor& t, int64_t start_row, int64_t end_row,
+           int64_t dim) {
           at::Tensor out;
           {
             pybind11::gil_scoped_release release;
-            out = tdx::materializeTensorShard(t, start_row, end_row);
+            out = tdx::materializeTensorShard(t, start_row, end_row, dim);
           }
           return out;
         },
-        "Materializes rows [start_row, end_row) of the deferred tensor's "
-        "dim 0 alone, bitwise-equal to that slice of a full native "
+        pybind11::arg("tensor"), pybind11::arg("start_row"),
+        pybind11::arg("end_row"), pybind11::arg("dim") = 0,
+        "Materializes indices [start_row, end_row) of the deferred tensor "
+        "along `dim` alone, bitwise-equal to that slice of a full native "
         "materialization (see docs/distributed_materialization.md).");
 }

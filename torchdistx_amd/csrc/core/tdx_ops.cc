@@ -13,6 +13,7 @@
 
 #include <cmath>
 #include <optional>
+#include <type_traits>
 
 #include <ATen/ATen.h>
 #include <torch/library.h>
@@ -231,6 +232,79 @@ at::Tensor& cpu_bernoulli_shard_(at::Tensor& shard, int64_t start,
   return shard;
 }
 
+// Windowed shard (any-dim slices): the shard is n_blocks contiguous
+// global ranges of block_len elements, block r starting at global
+// element g_off + r * g_stride — `full.narrow(dim, s, len)` flattened.
+// Same counter layout as the flat range, so each block is bitwise the
+// matching stretch of the full tensor (CUDA twin:
+// csrc/hip/init_kernels.hip rng_shard_window_kernel).
+template <RngKind kKind>
+void cpuPhiloxWindow(at::Tensor& shard, int64_t n_blocks, int64_t block_len,
+                     int64_t g_stride, int64_t g_off, float a, float b,
+                     uint64_t seed, uint64_t offset) {
+  TORCH_CHECK(shard.is_contiguous(),
+              "tdx CPU init requires contiguous tensors");
+  TORCH_CHECK(n_blocks >= 0 && block_len >= 0 && g_stride >= block_len &&
+                  g_off >= 0,
+              "invalid shard window");
+  TORCH_CHECK(shard.numel() == n_blocks * block_len,
+              "shard numel must equal n_blocks * block_len");
+  auto run = [&](auto* out) {
+    using T = std::remove_pointer_t<decltype(out)>;
+    for (int64_t r = 0; r < n_blocks; ++r) {
+      const int64_t lo = g_off + r * g_stride;
+      cpuPhiloxRange<T, kKind>(out + r * block_len, lo, lo + block_len, a,
+                               b, seed, offset);
+    }
+  };
+  switch (shard.scalar_type()) {
+    case at::kFloat:
+      run(shard.data_ptr<float>());
+      break;
+    case at::kBFloat16:
+      run(shard.data_ptr<at::BFloat16>());
+      break;
+    case at::kHalf:
+      run(shard.data_ptr<at::Half>());
+      break;
+    default:
+      TORCH_CHECK(false, "tdx CPU init supports float32/bf16/fp16, got ",
+                  shard.scalar_type());
+  }
+}
+
+at::Tensor& cpu_uniform_shard_win_(at::Tensor& shard, int64_t n_blocks,
+                                   int64_t block_len, int64_t g_stride,
+                                   int64_t g_off, double from, double to,
+                                   int64_t seed, int64_t offset) {
+  cpuPhiloxWindow<RngKind::kUniform>(
+      shard, n_blocks, block_len, g_stride, g_off, static_cast<float>(from),
+      static_cast<float>(to - from), static_cast<uint64_t>(seed),
+      static_cast<uint64_t>(offset));
+  return shard;
+}
+
+at::Tensor& cpu_normal_shard_win_(at::Tensor& shard, int64_t n_blocks,
+                                  int64_t block_len, int64_t g_stride,
+                                  int64_t g_off, double mean, double std,
+                                  int64_t seed, int64_t offset) {
+  cpuPhiloxWindow<RngKind::kNormal>(
+      shard, n_blocks, block_len, g_stride, g_off, static_cast<float>(mean),
+      static_cast<float>(std), static_cast<uint64_t>(seed),
+      static_cast<uint64_t>(offset));
+  return shard;
+}
+
+at::Tensor& cpu_bernoulli_shard_win_(at::Tensor& shard, int64_t n_blocks,
+                                     int64_t block_len, int64_t g_stride,
+                                     int64_t g_off, double p, int64_t seed,
+                                     int64_t offset) {
+  cpuPhiloxWindow<RngKind::kBernoulli>(
+      shard, n_blocks, block_len, g_stride, g_off, static_cast<float>(p),
+      0.0f, static_cast<uint64_t>(seed), static_cast<uint64_t>(offset));
+  return shard;
+}
+
 TORCH_LIBRARY(tdx, m) {
   m.def(
       "uniform_(Tensor(a!) self, float from=0., float to=1., *, "
@@ -258,6 +332,18 @@ TORCH_LIBRARY(tdx, m) {
   m.def(
       "bernoulli_shard_(Tensor(a!) shard, int start, int end, float p=0.5, "
       "*, int seed, int offset) -> Tensor(a!)");
+  m.def(
+      "uniform_shard_win_(Tensor(a!) shard, int n_blocks, int block_len, "
+      "int g_stride, int g_off, float from=0., float to=1., *, int seed, "
+      "int offset) -> Tensor(a!)");
+  m.def(
+      "normal_shard_win_(Tensor(a!) shard, int n_blocks, int block_len, "
+      "int g_stride, int g_off, float mean=0., float std=1., *, int seed, "
+      "int offset) -> Tensor(a!)");
+  m.def(
+      "bernoulli_shard_win_(Tensor(a!) shard, int n_blocks, int block_len, "
+      "int g_stride, int g_off, float p=0.5, *, int seed, int offset) "
+      "-> Tensor(a!)");
 }
 
 TORCH_LIBRARY_IMPL(tdx, CPU, m) {
@@ -270,6 +356,9 @@ TORCH_LIBRARY_IMPL(tdx, CPU, m) {
   m.impl("normal_shard_", cpu_normal_shard_);
   m.impl("bernoulli_", cpu_bernoulli_);
   m.impl("bernoulli_shard_", cpu_bernoulli_shard_);
+  m.impl("uniform_shard_win_", cpu_uniform_shard_win_);
+  m.impl("normal_shard_win_", cpu_normal_shard_win_);
+  m.impl("bernoulli_shard_win_", cpu_bernoulli_shard_win_);
 }
 
 }  // namespace

@@ -557,6 +557,150 @@ at::Tensor& tdx_bernoulli_shard_(at::Tensor& shard, int64_t start,
   return shard;
 }
 
+// Windowed shard kernel: materializes `full.narrow(dim, s, len)` of a
+// contiguous virtual full tensor for any dim — the slice is n_blocks
+// contiguous global ranges of block_len elements, block r starting at
+// global element g_off + r * g_stride (dim 0 is the n_blocks == 1
+// special case, served by rng_shard_kernel above). Shard flat index i
+// maps to global element
+//   g(i) = g_off + (i / block_len) * g_stride + (i % block_len),
+// and the value of global element g is lane g % kElems of Philox group
+// g / kElems — identical to the full-tensor kernel, so any-dim slices
+// are bitwise sub-tensors of the full materialization. When g_off,
+// block_len and g_stride are all group-multiples (the common case:
+// trailing dims of transformer weights are multiples of 8) every group
+// of the window is whole and the destination is 16-byte aligned, so the
+// fast path is group-indexed with vector stores like the flat kernel;
+// odd geometries store elementwise.
+template <typename T, Dist kDist>
+__global__ void rng_shard_window_kernel(T* __restrict__ out,
+                                        int64_t n_blocks,
+                                        int64_t block_len,
+                                        int64_t g_stride,
+                                        int64_t g_off,
+                                        float a,
+                                        float b,
+                                        uint64_t seed,
+                                        uint64_t offset) {
+  constexpr int kElems = VecTraits<T>::kElems;
+  using Vec = typename VecTraits<T>::Vec;
+  const int64_t stride = static_cast<int64_t>(gridDim.x) * blockDim.x;
+  const bool vec_aligned = (g_off % kElems) == 0 &&
+                           (block_len % kElems) == 0 &&
+                           (g_stride % kElems) == 0;
+  if (vec_aligned) {
+    const int64_t gpb = block_len / kElems;  // groups per window block
+    const int64_t n_groups = n_blocks * gpb;
+    const int64_t g0 = g_off / kElems;
+    const int64_t gs = g_stride / kElems;
+    for (int64_t j =
+             blockIdx.x * static_cast<int64_t>(blockDim.x) + threadIdx.x;
+         j < n_groups; j += stride) {
+      const int64_t g = g0 + (j / gpb) * gs + j % gpb;
+      float vals[kElems];
+      rngGroupValues<T, kDist>(static_cast<uint64_t>(g), a, b, seed, offset,
+                               vals);
+      Vec v;
+      T* vp = reinterpret_cast<T*>(&v);
+#pragma unroll
+      for (int e = 0; e < kElems; ++e) {
+        vp[e] = from_float<T>(vals[e]);
+      }
+      *reinterpret_cast<Vec*>(out + j * kElems) = v;
+    }
+    return;
+  }
+  const int64_t n = n_blocks * block_len;
+  for (int64_t i =
+           blockIdx.x * static_cast<int64_t>(blockDim.x) + threadIdx.x;
+       i < n; i += stride) {
+    const int64_t ge = g_off + (i / block_len) * g_stride + i % block_len;
+    const int64_t g = ge / kElems;
+    float vals[kElems];
+    rngGroupValues<T, kDist>(static_cast<uint64_t>(g), a, b, seed, offset,
+                             vals);
+    out[i] = from_float<T>(vals[ge - g * kElems]);
+  }
+}
+
+template <Dist kDist>
+void launchRngShardWindow(at::Tensor& shard, int64_t n_blocks,
+                          int64_t block_len, int64_t g_stride, int64_t g_off,
+                          double p0, double p1, int64_t seed,
+                          int64_t offset) {
+  TORCH_CHECK(shard.is_contiguous(),
+              "tdx shard kernels require contiguous tensors");
+  TORCH_CHECK(n_blocks >= 0 && block_len >= 0 && g_stride >= block_len &&
+                  g_off >= 0,
+              "invalid shard window");
+  TORCH_CHECK(shard.numel() == n_blocks * block_len,
+              "shard numel must equal n_blocks * block_len");
+  if (shard.numel() == 0) {
+    return;
+  }
+  float a = static_cast<float>(p0);
+  float b = kDist == Dist::kUniform ? static_cast<float>(p1 - p0)
+                                    : static_cast<float>(p1);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  auto launch = [&](auto type_tag) {
+    using T = decltype(type_tag);
+    const int64_t n_work =
+        (shard.numel() + VecTraits<T>::kElems - 1) / VecTraits<T>::kElems;
+    hipLaunchKernelGGL((rng_shard_window_kernel<T, kDist>),
+                       dim3(numBlocks(n_work)), dim3(kBlock), 0,
+                       stream.stream(),
+                       reinterpret_cast<T*>(shard.data_ptr()), n_blocks,
+                       block_len, g_stride, g_off, a, b,
+                       static_cast<uint64_t>(seed),
+                       static_cast<uint64_t>(offset));
+    C10_HIP_KERNEL_LAUNCH_CHECK();
+  };
+  switch (shard.scalar_type()) {
+    case at::kFloat:
+      launch(float{});
+      break;
+    case at::kBFloat16:
+      launch(__hip_bfloat16{});
+      break;
+    case at::kHalf:
+      launch(__half{});
+      break;
+    default:
+      TORCH_CHECK(false, "tdx shard kernels support float32/bf16/fp16, got ",
+                  shard.scalar_type());
+  }
+}
+
+at::Tensor& tdx_uniform_shard_win_(at::Tensor& shard, int64_t n_blocks,
+                                   int64_t block_len, int64_t g_stride,
+                                   int64_t g_off, double from, double to,
+                                   int64_t seed, int64_t offset) {
+  launchRngShardWindow<Dist::kUniform>(shard, n_blocks, block_len, g_stride,
+                                       g_off, from, to, seed, offset);
+  return shard;
+}
+
+at::Tensor& tdx_normal_shard_win_(at::Tensor& shard, int64_t n_blocks,
+                                  int64_t block_len, int64_t g_stride,
+                                  int64_t g_off, double mean, double std,
+                                  int64_t seed, int64_t offset) {
+  TORCH_CHECK(std >= 0.0, "normal_ expects std >= 0.0, but found std=", std);
+  launchRngShardWindow<Dist::kNormal>(shard, n_blocks, block_len, g_stride,
+                                      g_off, mean, std, seed, offset);
+  return shard;
+}
+
+at::Tensor& tdx_bernoulli_shard_win_(at::Tensor& shard, int64_t n_blocks,
+                                     int64_t block_len, int64_t g_stride,
+                                     int64_t g_off, double p, int64_t seed,
+                                     int64_t offset) {
+  TORCH_CHECK(0.0 <= p && p <= 1.0,
+              "bernoulli_ expects 0 <= p <= 1, but found p=", p);
+  launchRngShardWindow<Dist::kBernoulli>(shard, n_blocks, block_len, g_stride,
+                                         g_off, p, 0.0, seed, offset);
+  return shard;
+}
+
 at::Tensor& tdx_fill_(at::Tensor& self, const at::Scalar& value) {
   TORCH_CHECK(self.is_contiguous(),
               "tdx init kernels require contiguous tensors");
@@ -949,6 +1093,9 @@ TORCH_LIBRARY_IMPL(tdx, CUDA, m) {
   m.impl("normal_shard_", tdx_normal_shard_);
   m.impl("bernoulli_", tdx_bernoulli_);
   m.impl("bernoulli_shard_", tdx_bernoulli_shard_);
+  m.impl("uniform_shard_win_", tdx_uniform_shard_win_);
+  m.impl("normal_shard_win_", tdx_normal_shard_win_);
+  m.impl("bernoulli_shard_win_", tdx_bernoulli_shard_win_);
 }
 
 }  // namespace

@@ -7,5 +7,6 @@ from torchdistx_amd.parallel.sharded_materialize import (  # noqa: F401
     materialize_module_dim0_sharded,
     materialize_module_dtensor,
     materialize_module_distributed,
+    materialize_module_tp_sharded,
     materialize_tensor_shard,
 )

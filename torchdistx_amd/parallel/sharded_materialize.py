@@ -398,16 +398,18 @@ def _materialize_allgather(
 
 
 def materialize_tensor_shard(
-    tensor: torch.Tensor, start_row: int, end_row: int
+    tensor: torch.Tensor, start_row: int, end_row: int, dim: int = 0
 ) -> torch.Tensor:
-    """Materializes rows [start_row, end_row) of a deferred tensor's dim 0
-    without touching the rest — bitwise-equal to the corresponding slice of
-    a full native materialization, at shard cost. See
-    csrc/core/deferred_init.h (materializeTensorShard) for the supported
-    tape shapes; unsupported tapes raise, so callers can fall back to
-    `materialize_tensor` + slicing."""
+    """Materializes indices [start_row, end_row) of a deferred tensor
+    along ``dim`` without touching the rest — bitwise-equal to the
+    corresponding slice of a full native materialization, at shard cost.
+    dim 0 is the FSDP ``Shard(0)`` / column-parallel split; dim 1 the
+    row-parallel (Megatron-style TP) split. See csrc/core/deferred_init.h
+    (materializeTensorShard) for the supported tape shapes; unsupported
+    tapes raise, so callers can fall back to `materialize_tensor` +
+    slicing."""
     try:
-        shard = _C.materialize_tensor_shard(tensor, start_row, end_row)
+        shard = _C.materialize_tensor_shard(tensor, start_row, end_row, dim)
     except RuntimeError as e:
         if "slice materialization" not in str(e):
             raise
@@ -417,7 +419,10 @@ def materialize_tensor_shard(
         # covers, and cross-rank consistency still holds (every rank
         # replays the same tape).
         full = _C.materialize_tensor(tensor)
-        shard = full[start_row:end_row].clone() if full.dim() > 0 else full
+        if full.dim() > 0:
+            shard = full.narrow(dim, start_row, end_row - start_row).clone()
+        else:
+            shard = full
     if tensor.requires_grad:
         shard.requires_grad_(True)
     return shard
@@ -455,6 +460,52 @@ def materialize_module_dim0_sharded(
         end = (rank + 1) * rows // world_size
         shards[name] = materialize_tensor_shard(t, start, end)
     return shards
+
+
+def materialize_module_tp_sharded(
+    module: Module,
+    shard_dims: Dict[str, int],
+    rank: Optional[int] = None,
+    world_size: Optional[int] = None,
+) -> Dict[str, torch.Tensor]:
+    """Tensor-parallel init: each named parameter/buffer listed in
+    ``shard_dims`` (fully-qualified name -> split dim) is split
+    contiguously along that dim across ``world_size`` ranks and only this
+    rank's slice is materialized; names NOT listed are fully materialized
+    (bitwise-replicated on every rank, e.g. row-parallel biases and
+    norms). Zero communication. With the Megatron convention on
+    ``nn.Linear`` weights ([out, in]): column-parallel shards dim 0,
+    row-parallel shards dim 1.
+
+    Returns {fully-qualified tensor name -> local shard or full tensor}.
+    The module's own entries are left fake (they describe the full
+    tensors).
+    """
+    if rank is None:
+        rank = dist.get_rank() if dist.is_initialized() else 0
+    if world_size is None:
+        world_size = dist.get_world_size() if dist.is_initialized() else 1
+
+    named = list(module.named_parameters()) + list(module.named_buffers())
+    names = {name for name, _ in named}
+    unknown = set(shard_dims) - names
+    if unknown:
+        raise ValueError(
+            f"shard_dims names not found in the module: {sorted(unknown)}"
+        )
+    out: Dict[str, torch.Tensor] = {}
+    for name, t in named:
+        if not _C.can_materialize(t):
+            continue
+        dim = shard_dims.get(name)
+        if dim is None:
+            out[name] = _C.materialize_tensor(t)
+            continue
+        n = t.shape[dim] if t.dim() > 0 else 1
+        start = rank * n // world_size
+        end = (rank + 1) * n // world_size
+        out[name] = materialize_tensor_shard(t, start, end, dim)
+    return out
 
 
 def materialize_module_dtensor(
