@@ -290,6 +290,52 @@ def test_dtensor_materialization_matches_full() -> None:
     assert all(run_distributed(_dtensor_worker, 2))
 
 
+def _dtensor_tp_worker(rank, world):
+    from torch.distributed.device_mesh import init_device_mesh
+    from torch.distributed.tensor import Replicate, Shard
+
+    from torchdistx_amd import _C, deferred_init
+    from torchdistx_amd.deferred_init import materialize_module
+    from torchdistx_amd.parallel import materialize_module_dtensor
+
+    class Block(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.up = torch.nn.Linear(16, 32)
+            self.down = torch.nn.Linear(32, 16)
+
+    shard_dims = {"down.weight": 1, "down.bias": None}
+
+    _C.set_native_init_cpu(True)
+    try:
+        mesh = init_device_mesh("cpu", (world,))
+        torch.manual_seed(12)
+        m = deferred_init(Block)
+        dts = materialize_module_dtensor(m, mesh, shard_dims=shard_dims)
+
+        torch.manual_seed(12)
+        ref = deferred_init(Block)
+        materialize_module(ref)
+        refp = dict(ref.named_parameters())
+        ok = all(
+            torch.equal(dt.full_tensor(), refp[name].detach())
+            for name, dt in dts.items()
+        )
+        ok = ok and dts["down.weight"].placements == (Shard(1),)
+        ok = ok and dts["down.bias"].placements == (Replicate(),)
+        ok = ok and dts["up.weight"].placements == (Shard(0),)
+        return ok
+    finally:
+        _C.set_native_init_cpu(False)
+
+
+def test_dtensor_tp_placements_match_full() -> None:
+    # shard_dims overrides per tensor: Shard(1) for row-parallel weights,
+    # Replicate() (full bitwise copy, zero communication) for their
+    # biases; everything still reassembles to the native full model.
+    assert all(run_distributed(_dtensor_tp_worker, 2))
+
+
 def test_slice_falls_back_for_explicit_generator() -> None:
     # An RNG op recorded with an explicit generator cannot use the
     # counter-based shard path; the wrapper transparently falls back to the

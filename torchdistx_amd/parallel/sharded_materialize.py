@@ -511,18 +511,25 @@ def materialize_module_tp_sharded(
 def materialize_module_dtensor(
     module: Module,
     device_mesh,
+    shard_dims: Optional[Dict[str, int]] = None,
 ) -> Dict[str, "torch.Tensor"]:
     """FSDP2-era init: every parameter/buffer of the deferred ``module``
-    materializes directly as a ``DTensor`` sharded on dim 0 over the given
-    1-D device mesh — each rank slice-materializes only its local chunk
-    (torch.chunk split semantics, matching ``distribute_tensor``), so the
-    full model never exists on any device. Returns {fqn -> DTensor}."""
-    from torch.distributed.tensor import DTensor, Shard
+    materializes directly as a ``DTensor`` over the given 1-D device
+    mesh — each rank slice-materializes only its local chunk
+    (torch.chunk split semantics, matching ``distribute_tensor``), so
+    the full model never exists on any device. By default every tensor
+    shards on dim 0 (the FSDP2 ``fully_shard`` layout); ``shard_dims``
+    overrides the split dim per fully-qualified name (e.g. dim 1 for
+    row-parallel TP weights), with ``None`` meaning ``Replicate()`` —
+    the rank then fully materializes that tensor (bitwise-identical
+    everywhere, zero communication). Returns {fqn -> DTensor}."""
+    from torch.distributed.tensor import DTensor, Replicate, Shard
 
     if device_mesh.ndim != 1:
         raise ValueError("materialize_module_dtensor expects a 1-D mesh")
     world = device_mesh.size()
     rank = device_mesh.get_local_rank()
+    shard_dims = shard_dims or {}
 
     out: Dict[str, torch.Tensor] = {}
     for name, t in list(module.named_parameters()) + list(
@@ -530,13 +537,21 @@ def materialize_module_dtensor(
     ):
         if not _C.can_materialize(t):
             continue
-        rows = t.shape[0] if t.dim() > 0 else 1
-        slot = -(-rows // world)  # torch.chunk: ceil-size slots, short tail
-        start = min(rank * slot, rows)
-        end = min(start + slot, rows)
-        local = materialize_tensor_shard(t, start, end)
+        dim = shard_dims.get(name, 0)
+        if dim is None:
+            full = _C.materialize_tensor(t)
+            out[name] = DTensor.from_local(
+                full, device_mesh, [Replicate()], run_check=False,
+                shape=t.shape, stride=t.stride(),
+            )
+            continue
+        n = t.shape[dim] if t.dim() > 0 else 1
+        slot = -(-n // world)  # torch.chunk: ceil-size slots, short tail
+        start = min(rank * slot, n)
+        end = min(start + slot, n)
+        local = materialize_tensor_shard(t, start, end, dim)
         out[name] = DTensor.from_local(
-            local, device_mesh, [Shard(0)], run_check=False,
+            local, device_mesh, [Shard(dim)], run_check=False,
             shape=t.shape, stride=t.stride(),
         )
     return out
