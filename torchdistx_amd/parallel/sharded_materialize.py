@@ -408,6 +408,8 @@ def materialize_tensor_shard(
     (materializeTensorShard) for the supported tape shapes; unsupported
     tapes raise, so callers can fall back to `materialize_tensor` +
     slicing."""
+    if dim < 0:
+        dim += max(tensor.dim(), 1)
     try:
         shard = _C.materialize_tensor_shard(tensor, start_row, end_row, dim)
     except RuntimeError as e:
