@@ -403,6 +403,37 @@ def run_selftest(args, dist, device, rank, world, use_cuda) -> int:
             break
     checks.append(("slice-reassembly", ok, detail))
 
+    # 5. tp: concatenating every rank's dim-1 (row-parallel) slices of
+    # each 2-d weight reconstructs the same native reference bitwise —
+    # the windowed shard kernels' cross-rank contract.
+    from torchdistx_amd.parallel import materialize_tensor_shard
+
+    torch.manual_seed(4242)
+    m2 = deferred_init(build_model, cfg, device=device, dtype=dtype)
+    with native_path():
+        my2 = {}
+        for name, p in m2.named_parameters():
+            if p.dim() != 2:
+                continue
+            n_cols = p.shape[1]
+            a = rank * n_cols // world
+            b = (rank + 1) * n_cols // world
+            my2[name] = materialize_tensor_shard(p, a, b, dim=1).detach().cpu()
+    if world > 1:
+        gathered2 = [None] * world
+        dist.all_gather_object(gathered2, my2)
+    else:
+        gathered2 = [my2]
+    ok = True
+    detail = ""
+    for name in my2:
+        full = torch.cat([g[name] for g in gathered2], dim=1)
+        if not torch.equal(full, ref_params[name].detach().cpu()):
+            ok = False
+            detail = f"tp dim-1 reassembly mismatch: {name}"
+            break
+    checks.append(("tp-dim1-reassembly", ok, detail))
+
     failed = [c for c in checks if not c[1]]
     if rank == 0:
         print(json.dumps({
