@@ -572,10 +572,14 @@ at::Tensor& tdx_bernoulli_shard_(at::Tensor& shard, int64_t start,
 // of the window is whole and the destination is 16-byte aligned, so the
 // fast path is group-indexed with vector stores like the flat kernel;
 // odd geometries store elementwise.
-template <typename T, Dist kDist>
+// IdxT is the SHARD-side loop index type: 32-bit up to 2^31 shard
+// groups/elements (the flat-kernel design rule — real shards never pay
+// 64-bit divides, the one VALU-expensive op here); only the GLOBAL
+// group id widens to 64-bit, with a multiply, not a divide.
+template <typename T, Dist kDist, typename IdxT>
 __global__ void rng_shard_window_kernel(T* __restrict__ out,
-                                        int64_t n_blocks,
-                                        int64_t block_len,
+                                        IdxT n_blocks,
+                                        IdxT block_len,
                                         int64_t g_stride,
                                         int64_t g_off,
                                         float a,
@@ -584,41 +588,42 @@ __global__ void rng_shard_window_kernel(T* __restrict__ out,
                                         uint64_t offset) {
   constexpr int kElems = VecTraits<T>::kElems;
   using Vec = typename VecTraits<T>::Vec;
-  const int64_t stride = static_cast<int64_t>(gridDim.x) * blockDim.x;
+  const IdxT stride = static_cast<IdxT>(gridDim.x) * blockDim.x;
   const bool vec_aligned = (g_off % kElems) == 0 &&
                            (block_len % kElems) == 0 &&
                            (g_stride % kElems) == 0;
   if (vec_aligned) {
-    const int64_t gpb = block_len / kElems;  // groups per window block
-    const int64_t n_groups = n_blocks * gpb;
-    const int64_t g0 = g_off / kElems;
-    const int64_t gs = g_stride / kElems;
-    for (int64_t j =
-             blockIdx.x * static_cast<int64_t>(blockDim.x) + threadIdx.x;
+    const IdxT gpb = block_len / kElems;  // groups per window block
+    const IdxT n_groups = n_blocks * gpb;
+    const uint64_t g0 = static_cast<uint64_t>(g_off) / kElems;
+    const uint64_t gs = static_cast<uint64_t>(g_stride) / kElems;
+    for (IdxT j = blockIdx.x * static_cast<IdxT>(blockDim.x) + threadIdx.x;
          j < n_groups; j += stride) {
-      const int64_t g = g0 + (j / gpb) * gs + j % gpb;
+      const IdxT blk = j / gpb;
+      const IdxT within = j - blk * gpb;
+      const uint64_t g = g0 + static_cast<uint64_t>(blk) * gs + within;
       float vals[kElems];
-      rngGroupValues<T, kDist>(static_cast<uint64_t>(g), a, b, seed, offset,
-                               vals);
+      rngGroupValues<T, kDist>(g, a, b, seed, offset, vals);
       Vec v;
       T* vp = reinterpret_cast<T*>(&v);
 #pragma unroll
       for (int e = 0; e < kElems; ++e) {
         vp[e] = from_float<T>(vals[e]);
       }
-      *reinterpret_cast<Vec*>(out + j * kElems) = v;
+      *reinterpret_cast<Vec*>(out + static_cast<uint64_t>(j) * kElems) = v;
     }
     return;
   }
-  const int64_t n = n_blocks * block_len;
-  for (int64_t i =
-           blockIdx.x * static_cast<int64_t>(blockDim.x) + threadIdx.x;
+  const IdxT n = n_blocks * block_len;
+  for (IdxT i = blockIdx.x * static_cast<IdxT>(blockDim.x) + threadIdx.x;
        i < n; i += stride) {
-    const int64_t ge = g_off + (i / block_len) * g_stride + i % block_len;
-    const int64_t g = ge / kElems;
+    const IdxT blk = i / block_len;
+    const IdxT rem = i - blk * block_len;
+    const uint64_t ge = static_cast<uint64_t>(g_off) +
+                        static_cast<uint64_t>(blk) * g_stride + rem;
+    const uint64_t g = ge / kElems;
     float vals[kElems];
-    rngGroupValues<T, kDist>(static_cast<uint64_t>(g), a, b, seed, offset,
-                             vals);
+    rngGroupValues<T, kDist>(g, a, b, seed, offset, vals);
     out[i] = from_float<T>(vals[ge - g * kElems]);
   }
 }
@@ -646,14 +651,23 @@ void launchRngShardWindow(at::Tensor& shard, int64_t n_blocks,
     using T = decltype(type_tag);
     const int64_t n_work =
         (shard.numel() + VecTraits<T>::kElems - 1) / VecTraits<T>::kElems;
-    hipLaunchKernelGGL((rng_shard_window_kernel<T, kDist>),
-                       dim3(numBlocks(n_work)), dim3(kBlock), 0,
-                       stream.stream(),
-                       reinterpret_cast<T*>(shard.data_ptr()), n_blocks,
-                       block_len, g_stride, g_off, a, b,
-                       static_cast<uint64_t>(seed),
-                       static_cast<uint64_t>(offset));
-    C10_HIP_KERNEL_LAUNCH_CHECK();
+    auto launch_idx = [&](auto idx_tag) {
+      using IdxT = decltype(idx_tag);
+      hipLaunchKernelGGL((rng_shard_window_kernel<T, kDist, IdxT>),
+                         dim3(numBlocks(n_work)), dim3(kBlock), 0,
+                         stream.stream(),
+                         reinterpret_cast<T*>(shard.data_ptr()),
+                         static_cast<IdxT>(n_blocks),
+                         static_cast<IdxT>(block_len), g_stride, g_off, a,
+                         b, static_cast<uint64_t>(seed),
+                         static_cast<uint64_t>(offset));
+      C10_HIP_KERNEL_LAUNCH_CHECK();
+    };
+    if (shard.numel() <= std::numeric_limits<int32_t>::max()) {
+      launch_idx(uint32_t{});
+    } else {
+      launch_idx(int64_t{});
+    }
   };
   switch (shard.scalar_type()) {
     case at::kFloat:
