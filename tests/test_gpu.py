@@ -341,6 +341,41 @@ def test_any_dim_slice_matches_full_on_gpu() -> None:
                     )
 
 
+def test_windowed_slice_64bit_index_path() -> None:
+    # Shards above 2^31 elements take the 64-bit IdxT instantiation of
+    # the windowed kernel; both the aligned vector path and the odd
+    # elementwise fallback must stay bitwise sub-tensors of the full
+    # materialization across that threshold. (~14 GB HBM total.)
+    from torch.nn import Module, Parameter
+
+    from torchdistx_amd import _C, deferred_init
+
+    R, C = 67000, 36000  # full: 2.41e9 elems bf16
+
+    class M(Module):
+        def __init__(self):
+            super().__init__()
+            w = torch.empty(R, C, dtype=torch.bfloat16, device="cuda")
+            w.normal_(0.0, 1.0)
+            self.p = Parameter(w)
+
+    torch.manual_seed(64001)
+    full = _C.materialize_tensor(deferred_init(M).p).detach()
+    torch.manual_seed(64001)
+    part = deferred_init(M)
+    # 33000 cols -> 2.21e9-elem shard (> 2^31), 8-aligned: vector path.
+    s = _C.materialize_tensor_shard(part.p, 0, 33000, 1)
+    assert s.numel() > 2**31
+    assert torch.equal(s, full.narrow(1, 0, 33000))
+    del s
+    # 33001 cols: odd width -> elementwise fallback, still > 2^31.
+    s2 = _C.materialize_tensor_shard(part.p, 2000, 35001, 1)
+    assert s2.numel() > 2**31
+    assert torch.equal(s2, full.narrow(1, 2000, 33001))
+    del s2, full, part
+    torch.cuda.empty_cache()
+
+
 def test_tp_sharded_linear_reassembles_on_gpu() -> None:
     # Megatron-style TP init on GPU: column-parallel (dim 0) + row-parallel
     # (dim 1) slices of real Linear layers reassemble the full weights
