@@ -537,6 +537,39 @@ def test_tp_sharded_module() -> None:
         _C.set_native_init_cpu(False)
 
 
+def test_slice_empty_ranges_any_dim() -> None:
+    # world > dim size leaves some ranks with empty slices: shape must be
+    # right and reassembly must still be exact, on both dims.
+    from torch.nn import Module, Parameter
+
+    from torchdistx_amd import _C, deferred_init
+    from torchdistx_amd.parallel import materialize_tensor_shard
+
+    class M(Module):
+        def __init__(self):
+            super().__init__()
+            self.p = Parameter(torch.empty(3, 5).normal_())
+
+    _C.set_native_init_cpu(True)
+    try:
+        for dim in (0, 1):
+            torch.manual_seed(17)
+            full = _C.materialize_tensor(deferred_init(M).p).detach()
+            torch.manual_seed(17)
+            part = deferred_init(M)
+            world = 7
+            n = part.p.shape[dim]
+            parts = []
+            for r in range(world):
+                a, b = r * n // world, (r + 1) * n // world
+                s = materialize_tensor_shard(part.p, a, b, dim)
+                assert s.shape[dim] == b - a
+                parts.append(s.detach())
+            assert torch.equal(torch.cat(parts, dim=dim), full)
+    finally:
+        _C.set_native_init_cpu(False)
+
+
 def test_bernoulli_native_and_slice() -> None:
     # bernoulli_ joins the pinned-Philox op set: native CPU replay draws
     # from the pinned counters and the shard path reproduces any row
