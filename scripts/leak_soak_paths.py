@@ -28,6 +28,18 @@ os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
 os.environ.setdefault("MASTER_PORT", "29790")
 dist.init_process_group("nccl", rank=0, world_size=1)
 
+def _tp_windowed(m):
+    # Row-parallel windowed slices of every 2-d weight (rank 1 of 4):
+    # the result dict goes out of scope, so device memory must return
+    # to baseline like the module paths.
+    from torchdistx_amd.parallel import materialize_tensor_shard
+
+    for _, p in m.named_parameters():
+        if p.dim() == 2:
+            n = p.shape[1]
+            materialize_tensor_shard(p, n // 4, n // 2, dim=1)
+
+
 PATHS = {
     "sequential": materialize_module,
     "batched": materialize_module_batched,
@@ -35,6 +47,7 @@ PATHS = {
     "broadcast": lambda m: materialize_module_distributed(
         m, mode="broadcast"
     ),
+    "tp_windowed": _tp_windowed,
 }
 
 baseline = None
