@@ -533,6 +533,10 @@ def test_tp_sharded_module() -> None:
         m = deferred_init(Block)
         with pytest.raises(ValueError, match="not found in the module"):
             materialize_module_tp_sharded(m, {"nope.weight": 0})
+        torch.manual_seed(99)
+        m = deferred_init(Block)
+        with pytest.raises(ValueError, match="out of range"):
+            materialize_module_tp_sharded(m, {"up.bias": 1})
     finally:
         _C.set_native_init_cpu(False)
 

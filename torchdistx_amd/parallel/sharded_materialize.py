@@ -503,6 +503,11 @@ def materialize_module_tp_sharded(
         if dim is None:
             out[name] = _C.materialize_tensor(t)
             continue
+        if not -max(t.dim(), 1) <= dim < max(t.dim(), 1):
+            raise ValueError(
+                f"shard_dims[{name!r}] = {dim} is out of range for a "
+                f"{t.dim()}-d tensor"
+            )
         n = t.shape[dim] if t.dim() > 0 else 1
         start = rank * n // world_size
         end = (rank + 1) * n // world_size
