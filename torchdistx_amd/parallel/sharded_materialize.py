@@ -552,6 +552,8 @@ def materialize_module_dtensor(
                 shape=t.shape, stride=t.stride(),
             )
             continue
+        if dim < 0:
+            dim += max(t.dim(), 1)
         n = t.shape[dim] if t.dim() > 0 else 1
         slot = -(-n // world)  # torch.chunk: ceil-size slots, short tail
         start = min(rank * slot, n)
