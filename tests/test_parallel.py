@@ -541,6 +541,27 @@ def test_tp_sharded_module() -> None:
         _C.set_native_init_cpu(False)
 
 
+def test_shard_window_op_validation() -> None:
+    # The windowed shard ops reject malformed windows loudly and no-op
+    # on empty ones (the empty-per-rank-slice case).
+    import pytest
+
+    import torchdistx_amd  # noqa: F401  (registers tdx:: ops)
+
+    t = torch.empty(6, dtype=torch.float32)
+    with pytest.raises(RuntimeError, match="invalid shard window"):
+        torch.ops.tdx.uniform_shard_win_(
+            t, 2, 3, 2, 0, 0.0, 1.0, seed=1, offset=0
+        )  # g_stride < block_len: blocks would overlap
+    with pytest.raises(RuntimeError, match="n_blocks"):
+        torch.ops.tdx.uniform_shard_win_(
+            t, 2, 4, 8, 0, 0.0, 1.0, seed=1, offset=0
+        )  # numel != n_blocks * block_len
+    e = torch.empty(0, dtype=torch.float32)
+    torch.ops.tdx.normal_shard_win_(e, 0, 0, 0, 0, 0.0, 1.0, seed=1, offset=0)
+    assert e.numel() == 0
+
+
 def test_slice_empty_ranges_any_dim() -> None:
     # world > dim size leaves some ranks with empty slices: shape must be
     # right and reassembly must still be exact, on both dims.
